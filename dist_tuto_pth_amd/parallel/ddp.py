@@ -1,0 +1,162 @@
+"""Synchronous data parallelism (L5 core of the reference).
+
+``average_gradients`` reproduces train_dist.py:94-100 / tuto.md:309-315
+with the intended semantics: *no* exact-type guard (the committed
+``if type(param) is torch.Tensor`` makes the body dead code on modern
+PyTorch — SURVEY.md §2.5.2), one all-reduce per parameter, then divide
+by world size.
+
+``DistributedDataParallel`` is the production-shaped path the tutorial
+points to (tuto.md:216,320): gradients are packed into flat buckets,
+each bucket's all-reduce is launched on a *side HIP stream* as soon as
+its last gradient is accumulated, overlapping communication with the
+rest of backward; xGMI is point-to-point (7 x ~153 GB/s links), so
+bucket sizes default large enough to amortize per-collective launch
+cost while still giving overlap.
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+
+from .. import dist
+from ..dist import ReduceOp
+
+
+def average_gradients(model: torch.nn.Module, group=None):
+    """All-reduce every parameter gradient and divide by world size
+    (train_dist.py:94-100, unguarded per the paper listing
+    tuto.md:310-314)."""
+    size = float(dist.get_world_size(group))
+    for param in model.parameters():
+        if param.grad is not None:
+            dist.all_reduce(param.grad.data, op=ReduceOp.SUM, g=group)
+            param.grad.data /= size
+
+
+class _Bucket:
+    def __init__(self, params: List[torch.nn.Parameter], dtype, device):
+        self.params = params
+        self.numels = [p.numel() for p in params]
+        self.offsets = []
+        off = 0
+        for n in self.numels:
+            self.offsets.append(off)
+            off += n
+        self.flat = torch.zeros(off, dtype=dtype, device=device)
+        self.pending = 0
+        self.work = None
+
+    def reset(self):
+        self.pending = len(self.params)
+
+
+class DistributedDataParallel(torch.nn.Module):
+    """Bucketed, backward-overlapped gradient averaging.
+
+    Semantics match ``average_gradients`` (SUM then /world) but packed:
+    per bucket one flat all-reduce instead of one per tensor
+    (vs the reference's per-tensor blocking pattern, train_dist.py:99),
+    launched on a side stream keyed off per-parameter
+    post-accumulate hooks.  ``finish_gradients()`` must run before
+    ``optimizer.step()`` (it syncs the side stream and scatters averaged
+    grads back).
+    """
+
+    def __init__(self, module: torch.nn.Module, bucket_cap_mb: float = 25.0,
+                 group=None):
+        super().__init__()
+        self.module = module
+        self.group = group
+        self.world = dist.get_world_size(group)
+        params = [p for p in module.parameters() if p.requires_grad]
+        self._params = params
+        device = params[0].device if params else torch.device("cpu")
+        self._use_stream = device.type == "cuda"
+        self._comm_stream = torch.cuda.Stream() if self._use_stream else None
+        self._events = []
+
+        # Bucket in reverse parameter order: backward completes gradients
+        # roughly output-to-input, so reverse order lets early buckets
+        # fire while backward is still running.
+        cap = int(bucket_cap_mb * 1024 * 1024)
+        self.buckets: List[_Bucket] = []
+        self._bucket_of = {}
+        cur, cur_bytes = [], 0
+        for p in reversed(params):
+            b = p.numel() * p.element_size()
+            if cur and cur_bytes + b > cap:
+                self._make_bucket(cur, device)
+                cur, cur_bytes = [], 0
+            cur.append(p)
+            cur_bytes += b
+        if cur:
+            self._make_bucket(cur, device)
+
+        for p in params:
+            bucket = self._bucket_of[p]
+            p.register_post_accumulate_grad_hook(
+                self._make_hook(bucket))
+        self._reset()
+
+    def _make_bucket(self, params, device):
+        b = _Bucket(params, params[0].dtype, device)
+        self.buckets.append(b)
+        for p in params:
+            self._bucket_of[p] = b
+
+    def _reset(self):
+        for b in self.buckets:
+            b.reset()
+        self._events = []
+
+    def _make_hook(self, bucket: _Bucket):
+        def hook(param):
+            bucket.pending -= 1
+            if bucket.pending == 0:
+                self._launch(bucket)
+        return hook
+
+    def _launch(self, bucket: _Bucket):
+        if self.world == 1:
+            return
+        if self._use_stream:
+            ev = torch.cuda.Event()
+            ev.record(torch.cuda.current_stream())
+            with torch.cuda.stream(self._comm_stream):
+                self._comm_stream.wait_event(ev)
+                for p, off, n in zip(bucket.params, bucket.offsets,
+                                     bucket.numels):
+                    bucket.flat[off:off + n].copy_(p.grad.reshape(-1))
+                dist.all_reduce(bucket.flat, op=ReduceOp.SUM, g=self.group)
+                bucket.flat /= self.world
+                for p, off, n in zip(bucket.params, bucket.offsets,
+                                     bucket.numels):
+                    p.grad.reshape(-1).copy_(bucket.flat[off:off + n])
+                done = torch.cuda.Event()
+                done.record(self._comm_stream)
+                self._events.append(done)
+        else:
+            for p, off, n in zip(bucket.params, bucket.offsets,
+                                 bucket.numels):
+                bucket.flat[off:off + n].copy_(p.grad.reshape(-1))
+            dist.all_reduce(bucket.flat, op=ReduceOp.SUM, g=self.group)
+            bucket.flat /= self.world
+            for p, off, n in zip(bucket.params, bucket.offsets,
+                                 bucket.numels):
+                p.grad.reshape(-1).copy_(bucket.flat[off:off + n])
+
+    def forward(self, *args, **kwargs):
+        self._reset()
+        return self.module(*args, **kwargs)
+
+    def finish_gradients(self):
+        """Block the main stream on all bucket all-reduces.  Call after
+        ``loss.backward()`` and before ``optimizer.step()``."""
+        if self._use_stream:
+            cur = torch.cuda.current_stream()
+            for ev in self._events:
+                cur.wait_event(ev)
+        self._events = []

@@ -1,0 +1,71 @@
+"""End-to-end distributed training tests (the train_dist.py:103-127 loop).
+
+Correctness signals: per-rank losses decrease, and —
+the hard invariant of synchronous SGD with averaged gradients — the
+model replicas stay IDENTICAL across ranks (losses themselves differ
+slightly because each rank evaluates its own data shard)."""
+
+import json
+import os
+import tempfile
+
+import torch
+
+from dist_tuto_pth_amd import training
+from dist_tuto_pth_amd.dist.launcher import launch
+from dist_tuto_pth_amd.parallel import SyntheticMNIST
+
+
+def _save(rank, losses, model, prefix):
+    out = os.environ["_TRAIN_OUT"]
+    with open(os.path.join(out, f"{prefix}{rank}.json"), "w") as f:
+        json.dump(losses, f)
+    torch.save(model.state_dict(), os.path.join(out, f"{prefix}{rank}.pt"))
+
+
+def _fn_train(rank, size):
+    ds = SyntheticMNIST(n=512, seed=1234)
+    losses, model = training.run(rank, size, epochs=2, device="cpu",
+                                 dataset=ds, batch_size=128)
+    _save(rank, losses, model, "r")
+
+
+def _fn_train_ddp(rank, size):
+    ds = SyntheticMNIST(n=512, seed=1234)
+    losses, model = training.run(rank, size, epochs=2, device="cpu",
+                                 mode="ddp", dataset=ds, batch_size=128)
+    _save(rank, losses, model, "d")
+
+
+def _check(outdir, prefix, size):
+    all_losses = []
+    states = []
+    for r in range(size):
+        with open(os.path.join(outdir, f"{prefix}{r}.json")) as f:
+            all_losses.append(json.load(f))
+        states.append(torch.load(os.path.join(outdir, f"{prefix}{r}.pt")))
+    # replicas identical across ranks (averaged gradients keep them in
+    # lockstep from the shared seed-1234 init, train_dist.py:105)
+    for r in range(1, size):
+        for k in states[0]:
+            assert torch.allclose(states[0][k], states[r][k],
+                                  atol=1e-6), k
+    # per-rank losses in the same ballpark and improving
+    for r in range(size):
+        assert all_losses[r][-1] < all_losses[r][0] + 1e-6
+    for a, b in zip(all_losses[0], all_losses[1]):
+        assert abs(a - b) < 0.2
+
+
+def test_sync_sgd_average_gradients_world2():
+    with tempfile.TemporaryDirectory() as d:
+        os.environ["_TRAIN_OUT"] = d
+        launch(_fn_train, 2, timeout=300)
+        _check(d, "r", 2)
+
+
+def test_sync_sgd_ddp_world2():
+    with tempfile.TemporaryDirectory() as d:
+        os.environ["_TRAIN_OUT"] = d
+        launch(_fn_train_ddp, 2, timeout=300)
+        _check(d, "d", 2)
