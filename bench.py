@@ -1,0 +1,155 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: ConvNet distributed synchronous-SGD training
+(BASELINE.json config 3 — "ConvNet on synthetic 28x28 MNIST-shaped data,
+DP with average_gradients(), NxMI355X"), metric = whole-job samples/sec.
+
+Launch (driver contract):
+  python bench.py --gpus 1 --steps K --warmup W
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+One rank per GPU over the native RCCL backend.  Weak scaling: per-GPU
+batch fixed at --batch (default 128, the reference's global batch,
+train_dist.py:85), so global batch = 128*N.  Synthetic data, random-init
+weights (no network in this environment), fp32 (the reference's dtype).
+Timing: W untimed warmup steps, barrier+sync, exactly K timed steps,
+barrier+sync, MAX elapsed over ranks; rank 0 prints one JSON line.
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from dist_tuto_pth_amd import dist, ops  # noqa: E402
+from dist_tuto_pth_amd.models import Net  # noqa: E402
+from dist_tuto_pth_amd.optim import FusedSGD  # noqa: E402
+from dist_tuto_pth_amd.parallel import average_gradients  # noqa: E402
+from dist_tuto_pth_amd.parallel.ddp import DistributedDataParallel  # noqa: E402
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=50)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--batch", type=int, default=128,
+                   help="per-GPU batch (weak scaling)")
+    p.add_argument("--mode", choices=["average_gradients", "ddp"],
+                   default="average_gradients")
+    p.add_argument("--graph", action="store_true",
+                   help="capture the training step in a hipGraph")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", args.gpus))
+    rank = int(os.environ.get("RANK", 0))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+
+    if world > 1:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29500")
+        dist.init_process_group("rccl", world_size=world, rank=rank,
+                                device_id=local_rank)
+    torch.cuda.set_device(local_rank)
+    device = f"cuda:{local_rank}"
+
+    torch.manual_seed(1234)
+    model = Net().to(device)
+    if world > 1:
+        for p in model.parameters():
+            dist.broadcast(p.data, src=0)
+    opt = FusedSGD(model.parameters(), lr=0.01, momentum=0.5,
+                   zero_grad_in_step=True)
+    ddp = DistributedDataParallel(model) if (args.mode == "ddp" and
+                                             world > 1) else None
+
+    g = torch.Generator(device="cpu").manual_seed(1234 + rank)
+    x = torch.randn(args.batch, 1, 28, 28, generator=g).to(device)
+    tgt = torch.randint(0, 10, (args.batch,), generator=g).to(device)
+
+    def step():
+        if ddp is not None:
+            loss = ops.nll_loss(ddp(x), tgt)
+            loss.backward()
+            ddp.finish_gradients()
+        else:
+            loss = ops.log_softmax_nll(model.forward_logits(x), tgt)
+            loss.backward()
+            if world > 1:
+                average_gradients(model)
+        opt.step()
+        return loss
+
+    def barrier_sync():
+        if world > 1:
+            dist.barrier()
+        torch.cuda.synchronize()
+
+    graph = None
+    if args.graph:
+        # warm once to settle allocator, then capture the whole step
+        for _ in range(3):
+            step()
+        torch.cuda.synchronize()
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            step()
+
+        def step():  # noqa: F811
+            graph.replay()
+
+    for _ in range(args.warmup):
+        step()
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    if world > 1:
+        t = torch.tensor([elapsed], device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    samples_per_sec = args.batch * world * args.steps / elapsed
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "ConvNet samples/sec",
+            "value": samples_per_sec,
+            "unit": "samples/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "Net-21840",
+                "global_batch": args.batch * world,
+                "input": "1x28x28",
+                "parallelism": f"dp{world}",
+                "grad_sync": args.mode,
+                "graph": bool(args.graph),
+            },
+        }))
+
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

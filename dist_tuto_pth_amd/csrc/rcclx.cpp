@@ -1,0 +1,532 @@
+// rcclx — native RCCL-over-xGMI backend for dist_tuto_pth_amd.
+//
+// Owns the layer the reference tutorial inherits from PyTorch 0.x THD
+// (tuto.md:404-419): a TCP key-value store for rendezvous (the
+// master/worker handshake of tuto.md:409-418 — rank 0 hosts, workers
+// connect and exchange through it), ncclUniqueId broadcast,
+// ncclCommInitRank (one rank per MI355X), the six collectives
+// (tuto.md:197-202), point-to-point send/recv (tuto.md:87-112), the four
+// reduce ops (tuto.md:190-193) and sub-communicators via ncclCommSplit
+// (tuto.md:182-184).
+//
+// Pure HIP + RCCL + POSIX sockets; no torch headers (tensors arrive as
+// raw device pointers + dtype/count from the Python wrapper), no CUDA
+// compatibility paths.  Build: hipcc --offload-arch=gfx950 (build.py).
+
+#include <hip/hip_runtime.h>
+#include <rccl/rccl.h>
+
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <arpa/inet.h>
+#include <netdb.h>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
+#include <sys/socket.h>
+#include <unistd.h>
+
+#include <atomic>
+#include <chrono>
+#include <condition_variable>
+#include <cstring>
+#include <map>
+#include <mutex>
+#include <stdexcept>
+#include <string>
+#include <thread>
+#include <vector>
+
+namespace py = pybind11;
+
+#define HIP_CHECK(cmd)                                                        \
+  do {                                                                        \
+    hipError_t e_ = (cmd);                                                    \
+    if (e_ != hipSuccess)                                                     \
+      throw std::runtime_error(std::string("HIP error: ") +                   \
+                               hipGetErrorString(e_) + " @ " #cmd);           \
+  } while (0)
+
+#define NCCL_CHECK(cmd)                                                       \
+  do {                                                                        \
+    ncclResult_t r_ = (cmd);                                                  \
+    if (r_ != ncclSuccess)                                                    \
+      throw std::runtime_error(std::string("RCCL error: ") +                  \
+                               ncclGetErrorString(r_) + " @ " #cmd);          \
+  } while (0)
+
+// ---------------------------------------------------------------------------
+// TCP store: the rendezvous the tutorial describes in prose
+// (tuto.md:409-419) — rank 0 is the master holding a key/value table;
+// every rank (incl. 0) connects as a client.  Wire protocol:
+//   SET : u8(1) u32 klen key u32 vlen val            -> u8(1)
+//   GET : u8(2) u32 klen key u32 timeout_ms          -> u32 vlen val
+//         (vlen == 0xFFFFFFFF signals timeout)
+//   ADD : u8(3) u32 klen key i64 delta               -> i64 new_value
+// ---------------------------------------------------------------------------
+namespace {
+
+void send_all(int fd, const void* buf, size_t n) {
+  const char* p = static_cast<const char*>(buf);
+  while (n) {
+    ssize_t k = ::send(fd, p, n, MSG_NOSIGNAL);
+    if (k <= 0) throw std::runtime_error("store: send failed");
+    p += k;
+    n -= static_cast<size_t>(k);
+  }
+}
+
+bool recv_all(int fd, void* buf, size_t n) {
+  char* p = static_cast<char*>(buf);
+  while (n) {
+    ssize_t k = ::recv(fd, p, n, 0);
+    if (k <= 0) return false;
+    p += k;
+    n -= static_cast<size_t>(k);
+  }
+  return true;
+}
+
+}  // namespace
+
+class StoreServer {
+ public:
+  StoreServer(const std::string& host, int port, int expected_clients)
+      : stop_(false) {
+    listen_fd_ = ::socket(AF_INET, SOCK_STREAM, 0);
+    if (listen_fd_ < 0) throw std::runtime_error("store: socket() failed");
+    int one = 1;
+    setsockopt(listen_fd_, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+    sockaddr_in addr{};
+    addr.sin_family = AF_INET;
+    addr.sin_port = htons(static_cast<uint16_t>(port));
+    addr.sin_addr.s_addr = INADDR_ANY;
+    if (::bind(listen_fd_, reinterpret_cast<sockaddr*>(&addr), sizeof(addr)))
+      throw std::runtime_error("store: bind failed on port " +
+                               std::to_string(port));
+    if (::listen(listen_fd_, 128))
+      throw std::runtime_error("store: listen failed");
+    accept_thread_ = std::thread([this] { accept_loop(); });
+    (void)expected_clients;
+  }
+
+  ~StoreServer() {
+    stop_ = true;
+    cv_.notify_all();  // unblock pending GET waits
+    ::shutdown(listen_fd_, SHUT_RDWR);
+    ::close(listen_fd_);
+    {
+      std::lock_guard<std::mutex> lk(threads_mu_);
+      for (int fd : client_fds_) ::shutdown(fd, SHUT_RDWR);
+    }
+    if (accept_thread_.joinable()) accept_thread_.join();
+    for (auto& t : client_threads_)
+      if (t.joinable()) t.join();
+  }
+
+ private:
+  void accept_loop() {
+    while (!stop_) {
+      int fd = ::accept(listen_fd_, nullptr, nullptr);
+      if (fd < 0) {
+        if (stop_) break;
+        continue;
+      }
+      int one = 1;
+      setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+      std::lock_guard<std::mutex> lk(threads_mu_);
+      client_fds_.push_back(fd);
+      client_threads_.emplace_back([this, fd] { serve(fd); });
+    }
+  }
+
+  void serve(int fd) {
+    for (;;) {
+      uint8_t op;
+      if (!recv_all(fd, &op, 1)) break;
+      uint32_t klen;
+      if (!recv_all(fd, &klen, 4)) break;
+      std::string key(klen, '\0');
+      if (!recv_all(fd, key.data(), klen)) break;
+      if (op == 1) {  // SET
+        uint32_t vlen;
+        if (!recv_all(fd, &vlen, 4)) break;
+        std::vector<char> val(vlen);
+        if (vlen && !recv_all(fd, val.data(), vlen)) break;
+        {
+          std::lock_guard<std::mutex> lk(mu_);
+          table_[key] = std::move(val);
+        }
+        cv_.notify_all();
+        uint8_t ok = 1;
+        send_all(fd, &ok, 1);
+      } else if (op == 2) {  // GET (blocking w/ timeout)
+        uint32_t timeout_ms;
+        if (!recv_all(fd, &timeout_ms, 4)) break;
+        std::vector<char> val;
+        bool found = false;
+        {
+          std::unique_lock<std::mutex> lk(mu_);
+          cv_.wait_for(
+              lk, std::chrono::milliseconds(timeout_ms), [&] {
+                return stop_ || table_.count(key) > 0;
+              });
+          found = table_.count(key) > 0;
+          if (found) val = table_[key];
+        }
+        uint32_t vlen = found ? static_cast<uint32_t>(val.size())
+                              : 0xFFFFFFFFu;
+        send_all(fd, &vlen, 4);
+        if (found && !val.empty()) send_all(fd, val.data(), val.size());
+      } else if (op == 3) {  // ADD
+        int64_t delta;
+        if (!recv_all(fd, &delta, 8)) break;
+        int64_t nv;
+        {
+          std::lock_guard<std::mutex> lk(mu_);
+          int64_t cur = 0;
+          auto it = table_.find(key);
+          if (it != table_.end() && it->second.size() == 8)
+            memcpy(&cur, it->second.data(), 8);
+          nv = cur + delta;
+          std::vector<char> val(8);
+          memcpy(val.data(), &nv, 8);
+          table_[key] = std::move(val);
+        }
+        cv_.notify_all();
+        send_all(fd, &nv, 8);
+      } else {
+        break;
+      }
+    }
+    ::close(fd);
+  }
+
+  int listen_fd_;
+  std::atomic<bool> stop_;
+  std::thread accept_thread_;
+  std::mutex threads_mu_;
+  std::vector<std::thread> client_threads_;
+  std::vector<int> client_fds_;
+  std::mutex mu_;
+  std::condition_variable cv_;
+  std::map<std::string, std::vector<char>> table_;
+};
+
+class TcpStore {
+ public:
+  TcpStore(const std::string& host, int port, int rank, int world,
+           bool is_server, int timeout_ms)
+      : timeout_ms_(timeout_ms) {
+    if (is_server)
+      server_.reset(new StoreServer(host, port, world));
+    // connect as client (with retries: workers may race the master,
+    // tuto.md:414 "workers wait for the master")
+    auto deadline = std::chrono::steady_clock::now() +
+                    std::chrono::milliseconds(timeout_ms);
+    for (;;) {
+      fd_ = ::socket(AF_INET, SOCK_STREAM, 0);
+      sockaddr_in addr{};
+      addr.sin_family = AF_INET;
+      addr.sin_port = htons(static_cast<uint16_t>(port));
+      hostent* he = gethostbyname(host.c_str());
+      if (he == nullptr)
+        throw std::runtime_error("store: cannot resolve " + host);
+      memcpy(&addr.sin_addr, he->h_addr_list[0], he->h_length);
+      if (::connect(fd_, reinterpret_cast<sockaddr*>(&addr),
+                    sizeof(addr)) == 0)
+        break;
+      ::close(fd_);
+      if (std::chrono::steady_clock::now() > deadline)
+        throw std::runtime_error("store: connect timed out to " + host +
+                                 ":" + std::to_string(port));
+      std::this_thread::sleep_for(std::chrono::milliseconds(50));
+    }
+    int one = 1;
+    setsockopt(fd_, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+  }
+
+  ~TcpStore() { ::close(fd_); }
+
+  void set(const std::string& key, const py::bytes& value) {
+    std::string v = value;
+    std::lock_guard<std::mutex> lk(mu_);
+    uint8_t op = 1;
+    uint32_t klen = key.size(), vlen = v.size();
+    send_all(fd_, &op, 1);
+    send_all(fd_, &klen, 4);
+    send_all(fd_, key.data(), klen);
+    send_all(fd_, &vlen, 4);
+    if (vlen) send_all(fd_, v.data(), vlen);
+    uint8_t ok;
+    if (!recv_all(fd_, &ok, 1)) throw std::runtime_error("store: set failed");
+  }
+
+  py::bytes get(const std::string& key) {
+    std::vector<char> val;
+    {
+      py::gil_scoped_release nogil;
+      std::lock_guard<std::mutex> lk(mu_);
+      uint8_t op = 2;
+      uint32_t klen = key.size(), to = timeout_ms_;
+      send_all(fd_, &op, 1);
+      send_all(fd_, &klen, 4);
+      send_all(fd_, key.data(), klen);
+      send_all(fd_, &to, 4);
+      uint32_t vlen;
+      if (!recv_all(fd_, &vlen, 4))
+        throw std::runtime_error("store: get failed");
+      if (vlen == 0xFFFFFFFFu)
+        throw std::runtime_error("store: get('" + key + "') timed out");
+      val.resize(vlen);
+      if (vlen && !recv_all(fd_, val.data(), vlen))
+        throw std::runtime_error("store: get payload failed");
+    }
+    return py::bytes(val.data(), val.size());
+  }
+
+  int64_t add(const std::string& key, int64_t delta) {
+    py::gil_scoped_release nogil;
+    std::lock_guard<std::mutex> lk(mu_);
+    uint8_t op = 3;
+    uint32_t klen = key.size();
+    send_all(fd_, &op, 1);
+    send_all(fd_, &klen, 4);
+    send_all(fd_, key.data(), klen);
+    send_all(fd_, &delta, 8);
+    int64_t nv;
+    if (!recv_all(fd_, &nv, 8)) throw std::runtime_error("store: add failed");
+    return nv;
+  }
+
+ private:
+  std::unique_ptr<StoreServer> server_;
+  int fd_;
+  int timeout_ms_;
+  std::mutex mu_;
+};
+
+// ---------------------------------------------------------------------------
+// RCCL communicator
+// ---------------------------------------------------------------------------
+static ncclDataType_t as_dtype(int d) { return static_cast<ncclDataType_t>(d); }
+static ncclRedOp_t as_op(int o) { return static_cast<ncclRedOp_t>(o); }
+
+class Comm {
+ public:
+  Comm(int nranks, int rank, const py::bytes& uid_bytes, int device)
+      : nranks_(nranks), rank_(rank), device_(device) {
+    std::string uid_s = uid_bytes;
+    if (uid_s.size() != sizeof(ncclUniqueId))
+      throw std::runtime_error("bad ncclUniqueId size");
+    ncclUniqueId uid;
+    memcpy(&uid, uid_s.data(), sizeof(uid));
+    py::gil_scoped_release nogil;
+    HIP_CHECK(hipSetDevice(device));
+    NCCL_CHECK(ncclCommInitRank(&comm_, nranks, uid, rank));
+  }
+
+  Comm(ncclComm_t c, int nranks, int rank, int device)
+      : comm_(c), nranks_(nranks), rank_(rank), device_(device) {}
+
+  void destroy() {
+    if (comm_) {
+      py::gil_scoped_release nogil;
+      ncclCommDestroy(comm_);
+      comm_ = nullptr;
+    }
+  }
+
+  Comm* split(int color, int key) {
+    ncclComm_t nc = nullptr;
+    {
+      py::gil_scoped_release nogil;
+      NCCL_CHECK(ncclCommSplit(comm_,
+                               color < 0 ? NCCL_SPLIT_NOCOLOR : color, key,
+                               &nc, nullptr));
+    }
+    if (color < 0) return nullptr;
+    int n, r;
+    NCCL_CHECK(ncclCommCount(nc, &n));
+    NCCL_CHECK(ncclCommUserRank(nc, &r));
+    return new Comm(nc, n, r, device_);
+  }
+
+  // ---- collectives (tuto.md:197-202) --------------------------------
+  void all_reduce(uintptr_t send, uintptr_t recv, size_t count, int dtype,
+                  int op, uintptr_t stream) {
+    py::gil_scoped_release nogil;
+    NCCL_CHECK(ncclAllReduce(reinterpret_cast<void*>(send),
+                             reinterpret_cast<void*>(recv), count,
+                             as_dtype(dtype), as_op(op), comm_,
+                             reinterpret_cast<hipStream_t>(stream)));
+  }
+
+  void broadcast(uintptr_t send, uintptr_t recv, size_t count, int dtype,
+                 int root, uintptr_t stream) {
+    py::gil_scoped_release nogil;
+    NCCL_CHECK(ncclBroadcast(reinterpret_cast<void*>(send),
+                             reinterpret_cast<void*>(recv), count,
+                             as_dtype(dtype), root, comm_,
+                             reinterpret_cast<hipStream_t>(stream)));
+  }
+
+  void reduce(uintptr_t send, uintptr_t recv, size_t count, int dtype,
+              int op, int root, uintptr_t stream) {
+    py::gil_scoped_release nogil;
+    NCCL_CHECK(ncclReduce(reinterpret_cast<void*>(send),
+                          reinterpret_cast<void*>(recv), count,
+                          as_dtype(dtype), as_op(op), root, comm_,
+                          reinterpret_cast<hipStream_t>(stream)));
+  }
+
+  void all_gather(uintptr_t send, uintptr_t recv, size_t sendcount,
+                  int dtype, uintptr_t stream) {
+    py::gil_scoped_release nogil;
+    NCCL_CHECK(ncclAllGather(reinterpret_cast<void*>(send),
+                             reinterpret_cast<void*>(recv), sendcount,
+                             as_dtype(dtype), comm_,
+                             reinterpret_cast<hipStream_t>(stream)));
+  }
+
+  void reduce_scatter(uintptr_t send, uintptr_t recv, size_t recvcount,
+                      int dtype, int op, uintptr_t stream) {
+    py::gil_scoped_release nogil;
+    NCCL_CHECK(ncclReduceScatter(reinterpret_cast<void*>(send),
+                                 reinterpret_cast<void*>(recv), recvcount,
+                                 as_dtype(dtype), as_op(op), comm_,
+                                 reinterpret_cast<hipStream_t>(stream)));
+  }
+
+  void gather(uintptr_t send, uintptr_t recv, size_t sendcount, int dtype,
+              int root, uintptr_t stream) {
+    py::gil_scoped_release nogil;
+    NCCL_CHECK(ncclGather(reinterpret_cast<void*>(send),
+                          reinterpret_cast<void*>(recv), sendcount,
+                          as_dtype(dtype), root, comm_,
+                          reinterpret_cast<hipStream_t>(stream)));
+  }
+
+  void scatter(uintptr_t send, uintptr_t recv, size_t recvcount, int dtype,
+               int root, uintptr_t stream) {
+    py::gil_scoped_release nogil;
+    NCCL_CHECK(ncclScatter(reinterpret_cast<void*>(send),
+                           reinterpret_cast<void*>(recv), recvcount,
+                           as_dtype(dtype), root, comm_,
+                           reinterpret_cast<hipStream_t>(stream)));
+  }
+
+  void all_to_all(uintptr_t send, uintptr_t recv, size_t count, int dtype,
+                  uintptr_t stream) {
+    py::gil_scoped_release nogil;
+    NCCL_CHECK(ncclAllToAll(reinterpret_cast<void*>(send),
+                            reinterpret_cast<void*>(recv), count,
+                            as_dtype(dtype), comm_,
+                            reinterpret_cast<hipStream_t>(stream)));
+  }
+
+  // ---- p2p (tuto.md:87-112; ring transport C5-C8) --------------------
+  void send(uintptr_t buf, size_t count, int dtype, int peer,
+            uintptr_t stream) {
+    py::gil_scoped_release nogil;
+    NCCL_CHECK(ncclSend(reinterpret_cast<void*>(buf), count,
+                        as_dtype(dtype), peer, comm_,
+                        reinterpret_cast<hipStream_t>(stream)));
+  }
+
+  void recv(uintptr_t buf, size_t count, int dtype, int peer,
+            uintptr_t stream) {
+    py::gil_scoped_release nogil;
+    NCCL_CHECK(ncclRecv(reinterpret_cast<void*>(buf), count,
+                        as_dtype(dtype), peer, comm_,
+                        reinterpret_cast<hipStream_t>(stream)));
+  }
+
+  void group_start() { NCCL_CHECK(ncclGroupStart()); }
+  void group_end() {
+    py::gil_scoped_release nogil;
+    NCCL_CHECK(ncclGroupEnd());
+  }
+
+  int rank() const { return rank_; }
+  int nranks() const { return nranks_; }
+  int device() const { return device_; }
+
+ private:
+  ncclComm_t comm_ = nullptr;
+  int nranks_, rank_, device_;
+};
+
+// ---------------------------------------------------------------------------
+// small HIP utilities for the Python side
+// ---------------------------------------------------------------------------
+static py::bytes get_unique_id() {
+  ncclUniqueId uid;
+  NCCL_CHECK(ncclGetUniqueId(&uid));
+  return py::bytes(reinterpret_cast<const char*>(&uid), sizeof(uid));
+}
+
+static uintptr_t record_event(uintptr_t stream) {
+  hipEvent_t ev;
+  HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+  HIP_CHECK(hipEventRecord(ev, reinterpret_cast<hipStream_t>(stream)));
+  return reinterpret_cast<uintptr_t>(ev);
+}
+
+static void event_wait(uintptr_t ev) {
+  py::gil_scoped_release nogil;
+  hipEvent_t e = reinterpret_cast<hipEvent_t>(ev);
+  HIP_CHECK(hipEventSynchronize(e));
+  HIP_CHECK(hipEventDestroy(e));
+}
+
+static void stream_sync(uintptr_t stream) {
+  py::gil_scoped_release nogil;
+  HIP_CHECK(hipStreamSynchronize(reinterpret_cast<hipStream_t>(stream)));
+}
+
+static int device_count() {
+  int n = 0;
+  if (hipGetDeviceCount(&n) != hipSuccess) return 0;
+  return n;
+}
+
+PYBIND11_MODULE(_rcclx, m) {
+  m.doc() = "native RCCL-over-xGMI backend (TCP store + communicator)";
+
+  py::class_<TcpStore>(m, "TcpStore")
+      .def(py::init<const std::string&, int, int, int, bool, int>(),
+           py::arg("host"), py::arg("port"), py::arg("rank"),
+           py::arg("world"), py::arg("is_server"),
+           py::arg("timeout_ms") = 300000)
+      .def("set", &TcpStore::set)
+      .def("get", &TcpStore::get)
+      .def("add", &TcpStore::add);
+
+  py::class_<Comm>(m, "Comm")
+      .def(py::init<int, int, const py::bytes&, int>(), py::arg("nranks"),
+           py::arg("rank"), py::arg("uid"), py::arg("device"))
+      .def("destroy", &Comm::destroy)
+      .def("split", &Comm::split, py::return_value_policy::take_ownership)
+      .def("all_reduce", &Comm::all_reduce)
+      .def("broadcast", &Comm::broadcast)
+      .def("reduce", &Comm::reduce)
+      .def("all_gather", &Comm::all_gather)
+      .def("reduce_scatter", &Comm::reduce_scatter)
+      .def("gather", &Comm::gather)
+      .def("scatter", &Comm::scatter)
+      .def("all_to_all", &Comm::all_to_all)
+      .def("send", &Comm::send)
+      .def("recv", &Comm::recv)
+      .def("group_start", &Comm::group_start)
+      .def("group_end", &Comm::group_end)
+      .def("rank", &Comm::rank)
+      .def("nranks", &Comm::nranks)
+      .def("device", &Comm::device);
+
+  m.def("get_unique_id", &get_unique_id);
+  m.def("record_event", &record_event);
+  m.def("event_wait", &event_wait);
+  m.def("stream_sync", &stream_sync);
+  m.def("device_count", &device_count);
+}

@@ -55,6 +55,20 @@ def _stream() -> int:
     return torch.cuda.current_stream().cuda_stream
 
 
+class _InactiveBackend:
+    """Placeholder held by ranks that are not members of a sub-group
+    (every op raises; mirrors torch.distributed's non-member groups)."""
+
+    def destroy(self):
+        pass
+
+    def __getattr__(self, name):
+        def _raise(*a, **k):
+            raise RuntimeError(
+                "this rank is not a member of the process group")
+        return _raise
+
+
 class _RcclBackend:
     """Backend implementation object held by a ProcessGroup.
 
@@ -99,11 +113,12 @@ class _RcclBackend:
         key = ranks.index(self._rank) if member else 0
         sub = self._comm.split(color, key)
         if not member:
-            return _RcclBackend(None, None, None, None, _comm=None,
-                                _store=self._store, _ws=0, _rk=-1)
-        return _RcclBackend(None, None, None, None, _comm=sub,
-                            _store=self._store, _ws=len(ranks),
-                            _rk=ranks.index(self._rank))
+            return _InactiveBackend()
+        b = _RcclBackend(None, None, None, None, _comm=sub,
+                         _store=self._store, _ws=len(ranks),
+                         _rk=ranks.index(self._rank))
+        b._barrier_buf = torch.zeros(1, device="cuda")
+        return b
 
     def destroy(self):
         if getattr(self, "_comm", None) is not None:
