@@ -1,0 +1,124 @@
+#!/usr/bin/env python3
+"""All-reduce bus-bandwidth benchmark.
+
+  config 2: 256 MB fp32 across N MI355X (the allreduce.py path)
+  config 5: bf16 1 GB (Llama-3-8B-layer-sized) at 1/2/4/8 GPUs
+
+Compares the RCCL built-in all-reduce against this package's hand-rolled
+algorithms (fullmesh / ring — algorithms/xgmi.py).  Bus bandwidth uses
+the standard nccl-tests convention: busBW = 2*(N-1)/N * bytes / time.
+
+Launch:
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 benchmarks/bench_allreduce.py \
+      --size-mb 256 --dtype fp32 --algos rccl,fullmesh,ring
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from dist_tuto_pth_amd import dist  # noqa: E402
+from dist_tuto_pth_amd.algorithms.xgmi import xgmi_all_reduce  # noqa: E402
+
+
+def run_algo(algo, t, iters, warmup, world):
+    def op():
+        if algo == "rccl":
+            dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        else:
+            xgmi_all_reduce(t, algo=algo)
+
+    for _ in range(warmup):
+        op()
+    if world > 1:
+        dist.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        op()
+    torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    el = (time.perf_counter() - t0) / iters
+    # max over ranks
+    if world > 1:
+        e = torch.tensor([el], device=t.device)
+        dist.all_reduce(e, op=dist.ReduceOp.MAX)
+        el = float(e.item())
+    bytes_ = t.numel() * t.element_size()
+    bus_bw = 2 * (world - 1) / world * bytes_ / el / 1e9 if world > 1 else 0.0
+    alg_bw = bytes_ / el / 1e9
+    return el, bus_bw, alg_bw
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--size-mb", type=float, default=256.0)
+    p.add_argument("--dtype", choices=["fp32", "bf16"], default="fp32")
+    p.add_argument("--iters", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--algos", default="rccl,fullmesh,ring")
+    p.add_argument("--check", action="store_true",
+                   help="verify results against RCCL first")
+    args = p.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    rank = int(os.environ.get("RANK", 0))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    if world > 1:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29500")
+        dist.init_process_group("rccl", world_size=world, rank=rank,
+                                device_id=local_rank)
+    torch.cuda.set_device(local_rank)
+    dtype = torch.float32 if args.dtype == "fp32" else torch.bfloat16
+    numel = int(args.size_mb * 1e6) // (4 if dtype == torch.float32 else 2)
+    g = torch.Generator().manual_seed(42 + rank)
+    base = torch.randn(numel, generator=g).to(dtype).cuda()
+
+    if args.check and world > 1:
+        ref = base.clone()
+        dist.all_reduce(ref, op=dist.ReduceOp.SUM)
+        for algo in ("fullmesh", "ring"):
+            t = base.clone()
+            xgmi_all_reduce(t, algo=algo)
+            torch.cuda.synchronize()
+            ok = torch.allclose(t.float(), ref.float(),
+                                atol=1e-2 if dtype == torch.bfloat16
+                                else 1e-4, rtol=1e-2)
+            if rank == 0:
+                print(f"# check {algo}: {'OK' if ok else 'MISMATCH'} "
+                      f"max|d|={float((t.float()-ref.float()).abs().max())}",
+                      flush=True)
+
+    results = {}
+    for algo in args.algos.split(","):
+        if world == 1 and algo != "rccl":
+            continue  # p2p algorithms need peers
+        t = base.clone()
+        el, bus, alg = run_algo(algo, t, args.iters, args.warmup, world)
+        results[algo] = {"ms": el * 1e3, "bus_GBps": bus, "alg_GBps": alg}
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "all-reduce bus BW (GB/s)",
+            "n_gpus": world,
+            "size_mb": args.size_mb,
+            "dtype": args.dtype,
+            "iters": args.iters,
+            "warmup": args.warmup,
+            "results": results,
+        }))
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -290,6 +290,14 @@ __global__ void relu_bwd_kernel(const float* __restrict__ gy,
 // no RNG-state tensor.  Channelwise variant draws one number per
 // (batch, channel) plane (Dropout2d, train_dist.py:60).
 // ===========================================================================
+// device-side seed bump: lets dropout RNG advance across hipGraph
+// replays (the seed lives in device memory; one tiny kernel increments
+// it before each dropout draw, stream-ordered).
+__global__ void bump_seed_kernel(unsigned long long* s) {
+  if (threadIdx.x == 0 && blockIdx.x == 0)
+    *s += 0x9E3779B97F4A7C15ull;
+}
+
 __device__ inline uint32_t mix32(uint64_t seed, uint64_t idx) {
   uint64_t z = seed + 0x9E3779B97F4A7C15ull * (idx + 1);
   z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
@@ -300,8 +308,10 @@ __device__ inline uint32_t mix32(uint64_t seed, uint64_t idx) {
 __global__ void dropout_fwd_kernel(const float* __restrict__ x,
                                    float* __restrict__ out,
                                    uint8_t* __restrict__ mask, int64_t n,
-                                   float p, float scale, uint64_t seed) {
+                                   float p, float scale,
+                                   const unsigned long long* __restrict__ sp) {
   const uint32_t thresh = (uint32_t)(p * 4294967296.0);
+  const uint64_t seed = sp[0];
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     uint32_t r = mix32(seed, (uint64_t)i) << 16 | (mix32(seed ^ 0xabcd, i) & 0xffff);
@@ -324,8 +334,10 @@ __global__ void dropout2d_fwd_kernel(const float* __restrict__ x,
                                      float* __restrict__ out,
                                      uint8_t* __restrict__ mask,
                                      int64_t planes, int64_t hw, float p,
-                                     float scale, uint64_t seed) {
+                                     float scale,
+                                     const unsigned long long* __restrict__ sp) {
   const uint32_t thresh = (uint32_t)(p * 4294967296.0);
+  const uint64_t seed = sp[0];
   const int64_t n = planes * hw;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
@@ -504,12 +516,16 @@ __global__ void nll_loss_fwd_kernel(const float* __restrict__ logp,
   if (threadIdx.x == 0) atomicAdd(loss, part[0] / B);
 }
 
+// gloss arrives as a DEVICE scalar pointer so the whole backward is
+// hipGraph-capturable (no host read of the grad seed).
 __global__ void nll_loss_bwd_kernel(const int64_t* __restrict__ target,
-                                    float* __restrict__ gx, float gloss,
+                                    float* __restrict__ gx,
+                                    const float* __restrict__ gloss,
                                     int B, int N) {
+  const float sc = -gloss[0] / B;
   for (int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; b < B;
        b += (int64_t)gridDim.x * blockDim.x)
-    gx[b * N + target[b]] = -gloss / B;
+    gx[b * N + target[b]] = sc;
 }
 
 // fused: logp + mean NLL in one pass (K9+K10)
@@ -545,8 +561,9 @@ __global__ void log_softmax_nll_fwd_kernel(const float* __restrict__ x,
 __global__ void log_softmax_nll_bwd_kernel(const float* __restrict__ logp,
                                            const int64_t* __restrict__ tgt,
                                            float* __restrict__ gx,
-                                           float gloss, int B, int N) {
-  const float sc = gloss / B;
+                                           const float* __restrict__ gloss,
+                                           int B, int N) {
+  const float sc = gloss[0] / B;
   for (int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; b < B;
        b += (int64_t)gridDim.x * blockDim.x) {
     const float* lr = logp + b * N;
@@ -637,6 +654,81 @@ __global__ void add_inplace_bf16_kernel(__hip_bfloat16* __restrict__ dst,
     dst[i] = __hadd(dst[i], src[i]);
 }
 
+// column reduction for the full-mesh reduce-scatter (K14): after the
+// grouped p2p exchange, rank i holds P-1 peer chunks contiguously in
+// scratch; one kernel folds them all into the owned chunk.
+// dst[j] (+)= sum_p src[p*stride + j].  fp32 accumulates fp32; bf16
+// accumulates in fp32 and rounds ONCE at the end (better than serial
+// bf16 adds — BASELINE config 5 accuracy note, SURVEY.md §7 hard part d).
+__global__ void reduce_columns_f32_kernel(float* __restrict__ dst,
+                                          const float* __restrict__ src,
+                                          int P, int64_t stride,
+                                          int64_t n, float scale) {
+  const int64_t n4 = n / 4;
+  float4* d4 = reinterpret_cast<float4*>(dst);
+  const float4* s4 = reinterpret_cast<const float4*>(src);
+  const int64_t stride4 = stride / 4;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float4 a = d4[i];
+    for (int p = 0; p < P; ++p) {
+      float4 b = s4[p * stride4 + i];
+      a.x += b.x; a.y += b.y; a.z += b.z; a.w += b.w;
+    }
+    a.x *= scale; a.y *= scale; a.z *= scale; a.w *= scale;
+    d4[i] = a;
+  }
+  for (int64_t i = n4 * 4 + blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float a = dst[i];
+    for (int p = 0; p < P; ++p) a += src[p * stride + i];
+    dst[i] = a * scale;
+  }
+}
+
+__global__ void reduce_columns_bf16_kernel(
+    __hip_bfloat16* __restrict__ dst,
+    const __hip_bfloat16* __restrict__ src, int P, int64_t stride,
+    int64_t n, float scale) {
+  const int64_t n8 = n / 8;
+  uint4* d8 = reinterpret_cast<uint4*>(dst);
+  const uint4* s8 = reinterpret_cast<const uint4*>(src);
+  const int64_t stride8 = stride / 8;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint4 av = d8[i];
+    const __hip_bfloat162* ah = reinterpret_cast<__hip_bfloat162*>(&av);
+    float2 acc[4];
+    #pragma unroll
+    for (int k = 0; k < 4; ++k) acc[k] = __bfloat1622float2(ah[k]);
+    for (int p = 0; p < P; ++p) {
+      uint4 bv = s8[p * stride8 + i];
+      const __hip_bfloat162* bh =
+          reinterpret_cast<const __hip_bfloat162*>(&bv);
+      #pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        float2 b = __bfloat1622float2(bh[k]);
+        acc[k].x += b.x;
+        acc[k].y += b.y;
+      }
+    }
+    uint4 ov;
+    __hip_bfloat162* oh = reinterpret_cast<__hip_bfloat162*>(&ov);
+    #pragma unroll
+    for (int k = 0; k < 4; ++k)
+      oh[k] = __float22bfloat162_rn(
+          make_float2(acc[k].x * scale, acc[k].y * scale));
+    d8[i] = ov;
+  }
+  for (int64_t i = n8 * 8 + blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float a = __bfloat162float(dst[i]);
+    for (int p = 0; p < P; ++p)
+      a += __bfloat162float(src[p * stride + i]);
+    dst[i] = __float2bfloat16(a * scale);
+  }
+}
+
 __global__ void scale_f32_kernel(float* __restrict__ dst, float s,
                                  int64_t n) {
   const int64_t n4 = n / 4;
@@ -722,11 +814,14 @@ void relu_bwd(uintptr_t gy, uintptr_t out, uintptr_t gx, int64_t n,
 }
 
 void dropout_fwd(uintptr_t x, uintptr_t out, uintptr_t mask, int64_t n,
-                 double p, uint64_t seed, uintptr_t stream) {
+                 double p, uintptr_t seed_dev, uintptr_t stream) {
   const float scale = 1.f / (1.f - (float)p);
+  hipLaunchKernelGGL(bump_seed_kernel, dim3(1), dim3(64), 0, S(stream),
+                     (unsigned long long*)seed_dev);
   hipLaunchKernelGGL(dropout_fwd_kernel, dim3(grid_for(n, BLK)), dim3(BLK),
                      0, S(stream), (const float*)x, (float*)out,
-                     (uint8_t*)mask, n, (float)p, scale, seed);
+                     (uint8_t*)mask, n, (float)p, scale,
+                     (const unsigned long long*)seed_dev);
 }
 
 void dropout_bwd(uintptr_t gy, uintptr_t mask, uintptr_t gx, int64_t n,
@@ -737,13 +832,16 @@ void dropout_bwd(uintptr_t gy, uintptr_t mask, uintptr_t gx, int64_t n,
 }
 
 void dropout2d_fwd(uintptr_t x, uintptr_t out, uintptr_t mask,
-                   int64_t planes, int64_t hw, double p, uint64_t seed,
-                   uintptr_t stream) {
+                   int64_t planes, int64_t hw, double p,
+                   uintptr_t seed_dev, uintptr_t stream) {
   const float scale = 1.f / (1.f - (float)p);
+  hipLaunchKernelGGL(bump_seed_kernel, dim3(1), dim3(64), 0, S(stream),
+                     (unsigned long long*)seed_dev);
   hipLaunchKernelGGL(dropout2d_fwd_kernel,
                      dim3(grid_for(planes * hw, BLK)), dim3(BLK), 0,
                      S(stream), (const float*)x, (float*)out,
-                     (uint8_t*)mask, planes, hw, (float)p, scale, seed);
+                     (uint8_t*)mask, planes, hw, (float)p, scale,
+                     (const unsigned long long*)seed_dev);
 }
 
 void dropout2d_bwd(uintptr_t gy, uintptr_t mask, uintptr_t gx,
@@ -811,11 +909,11 @@ void nll_loss_fwd(uintptr_t logp, uintptr_t target, uintptr_t loss, int B,
                      (const int64_t*)target, (float*)loss, B, N);
 }
 
-void nll_loss_bwd(uintptr_t target, uintptr_t gx, double gloss, int B,
+void nll_loss_bwd(uintptr_t target, uintptr_t gx, uintptr_t gloss, int B,
                   int N, uintptr_t stream) {
   hipLaunchKernelGGL(nll_loss_bwd_kernel, dim3(grid_for(B, BLK)), dim3(BLK),
                      0, S(stream), (const int64_t*)target, (float*)gx,
-                     (float)gloss, B, N);
+                     (const float*)gloss, B, N);
 }
 
 void log_softmax_nll_fwd(uintptr_t x, uintptr_t tgt, uintptr_t logp,
@@ -827,10 +925,11 @@ void log_softmax_nll_fwd(uintptr_t x, uintptr_t tgt, uintptr_t logp,
 }
 
 void log_softmax_nll_bwd(uintptr_t logp, uintptr_t tgt, uintptr_t gx,
-                         double gloss, int B, int N, uintptr_t stream) {
+                         uintptr_t gloss, int B, int N, uintptr_t stream) {
   hipLaunchKernelGGL(log_softmax_nll_bwd_kernel, dim3(grid_for(B, BLK)),
                      dim3(BLK), 0, S(stream), (const float*)logp,
-                     (const int64_t*)tgt, (float*)gx, (float)gloss, B, N);
+                     (const int64_t*)tgt, (float*)gx, (const float*)gloss,
+                     B, N);
 }
 
 void sgd_step(const std::vector<uintptr_t>& ps,
@@ -875,6 +974,23 @@ void add_inplace(uintptr_t dst, uintptr_t src, int64_t n, int dtype,
   }
 }
 
+void reduce_columns(uintptr_t dst, uintptr_t src, int P, int64_t stride,
+                    int64_t n, double scale, int dtype, uintptr_t stream) {
+  if (dtype == 7) {
+    hipLaunchKernelGGL(reduce_columns_f32_kernel,
+                       dim3(grid_for(n, BLK, 4)), dim3(BLK), 0, S(stream),
+                       (float*)dst, (const float*)src, P, stride, n,
+                       (float)scale);
+  } else if (dtype == 9) {
+    hipLaunchKernelGGL(reduce_columns_bf16_kernel,
+                       dim3(grid_for(n, BLK, 8)), dim3(BLK), 0, S(stream),
+                       (__hip_bfloat16*)dst, (const __hip_bfloat16*)src, P,
+                       stride, n, (float)scale);
+  } else {
+    throw std::runtime_error("reduce_columns: unsupported dtype");
+  }
+}
+
 void scale_f32(uintptr_t dst, double s, int64_t n, uintptr_t stream) {
   hipLaunchKernelGGL(scale_f32_kernel, dim3(grid_for(n, BLK, 4)), dim3(BLK),
                      0, S(stream), (float*)dst, (float)s, n);
@@ -904,5 +1020,6 @@ PYBIND11_MODULE(_kernels, m) {
   m.def("log_softmax_nll_bwd", &log_softmax_nll_bwd);
   m.def("sgd_step", &sgd_step);
   m.def("add_inplace", &add_inplace);
+  m.def("reduce_columns", &reduce_columns);
   m.def("scale_f32", &scale_f32);
 }

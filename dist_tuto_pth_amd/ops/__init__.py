@@ -163,12 +163,18 @@ class _Relu(torch.autograd.Function):
 # K5/K6 — dropout2d (per-channel mask, train_dist.py:60,66) and
 # elementwise dropout (train_dist.py:69), device-side philox-style RNG.
 # ---------------------------------------------------------------------------
-_dropout_seed = [12345]
+# per-device RNG seed counter living in DEVICE memory: the dropout
+# kernels bump-and-read it on the stream, so masks advance correctly
+# across hipGraph replays (host-side seeds would be frozen at capture).
+_seed_bufs = {}
 
 
-def _next_seed():
-    _dropout_seed[0] = (_dropout_seed[0] * 6364136223846793005 + 1442695040888963407) & 0xFFFFFFFFFFFFFFFF
-    return _dropout_seed[0]
+def _seed_ptr(device) -> int:
+    key = device.index
+    if key not in _seed_bufs:
+        _seed_bufs[key] = torch.tensor([12345], dtype=torch.int64,
+                                       device=device)
+    return _seed_bufs[key].data_ptr()
 
 
 class _Dropout(torch.autograd.Function):
@@ -188,13 +194,13 @@ class _Dropout(torch.autograd.Function):
                                    dtype=torch.uint8)
                 k.dropout2d_fwd(x.data_ptr(), out.data_ptr(),
                                 mask.data_ptr(), B * C, hw, p,
-                                _next_seed(), _stream())
+                                _seed_ptr(x.device), _stream())
             else:
                 mask = torch.empty(x.numel(), device=x.device,
                                    dtype=torch.uint8)
                 k.dropout_fwd(x.data_ptr(), out.data_ptr(),
                               mask.data_ptr(), x.numel(), p,
-                              _next_seed(), _stream())
+                              _seed_ptr(x.device), _stream())
             ctx.save_for_backward(mask)
             ctx.meta = (scale, channelwise, x.shape)
             ctx.mask = mask
@@ -356,8 +362,9 @@ class _NllLoss(torch.autograd.Function):
             k = _k()
             B, N = logp.shape
             gx = torch.zeros_like(logp)
+            gl = gloss.contiguous()
             k.nll_loss_bwd(target.data_ptr(), gx.data_ptr(),
-                           float(gloss), B, N, _stream())
+                           gl.data_ptr(), B, N, _stream())
             return gx, None
         B, N = logp.shape
         gx = torch.zeros_like(logp)
@@ -396,8 +403,9 @@ class _LogSoftmaxNll(torch.autograd.Function):
         if logp.is_cuda:
             k = _k()
             gx = torch.empty_like(logp)
+            gl = gloss.contiguous()
             k.log_softmax_nll_bwd(logp.data_ptr(), target.data_ptr(),
-                                  gx.data_ptr(), float(gloss), B, N,
+                                  gx.data_ptr(), gl.data_ptr(), B, N,
                                   _stream())
             return gx, None
         gx = logp.exp()
