@@ -44,6 +44,9 @@ def parse_args():
                    default="average_gradients")
     p.add_argument("--graph", action="store_true",
                    help="capture the training step in a hipGraph")
+    p.add_argument("--no-fused", action="store_true",
+                   help="use the modular per-op kernel pipeline instead "
+                        "of the fused whole-Net kernels")
     return p.parse_args()
 
 
@@ -53,13 +56,14 @@ def main():
     rank = int(os.environ.get("RANK", 0))
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
 
+    dev_idx = local_rank % torch.cuda.device_count()
     if world > 1:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29500")
         dist.init_process_group("rccl", world_size=world, rank=rank,
-                                device_id=local_rank)
-    torch.cuda.set_device(local_rank)
-    device = f"cuda:{local_rank}"
+                                device_id=dev_idx)
+    torch.cuda.set_device(dev_idx)
+    device = f"cuda:{dev_idx}"
 
     torch.manual_seed(1234)
     model = Net().to(device)
@@ -75,7 +79,23 @@ def main():
     x = torch.randn(args.batch, 1, 28, 28, generator=g).to(device)
     tgt = torch.randint(0, 10, (args.batch,), generator=g).to(device)
 
+    use_fused = not args.no_fused and args.mode == "average_gradients"
+    flat_grads = None
+    if use_fused:
+        from dist_tuto_pth_amd.ops.fused import (attach_flat_grads,
+                                                 net_fused_step)
+        flat_grads = attach_flat_grads(model)
+
     def step():
+        if use_fused:
+            loss = net_fused_step(model, x, tgt)
+            if world > 1:
+                # one flat all-reduce with built-in averaging: the
+                # semantics of average_gradients (train_dist.py:94-100)
+                # in a single xGMI message
+                dist.all_reduce(flat_grads, op=dist.ReduceOp.AVG)
+            opt.step()
+            return loss
         if ddp is not None:
             loss = ops.nll_loss(ddp(x), tgt)
             loss.backward()
@@ -143,6 +163,7 @@ def main():
                 "input": "1x28x28",
                 "parallelism": f"dp{world}",
                 "grad_sync": args.mode,
+                "fused": use_fused,
                 "graph": bool(args.graph),
             },
         }))

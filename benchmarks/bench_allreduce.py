@@ -72,12 +72,13 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", 1))
     rank = int(os.environ.get("RANK", 0))
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    dev_idx = local_rank % torch.cuda.device_count()
     if world > 1:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29500")
         dist.init_process_group("rccl", world_size=world, rank=rank,
-                                device_id=local_rank)
-    torch.cuda.set_device(local_rank)
+                                device_id=dev_idx)
+    torch.cuda.set_device(dev_idx)
     dtype = torch.float32 if args.dtype == "fp32" else torch.bfloat16
     numel = int(args.size_mb * 1e6) // (4 if dtype == torch.float32 else 2)
     g = torch.Generator().manual_seed(42 + rank)

@@ -63,42 +63,40 @@ __global__ void conv2d_fwd_kernel(const float* __restrict__ x,
                                   int K, int R, int S_) {
   extern __shared__ __attribute__((aligned(16))) float smem[];
   const int OH = H - R + 1, OW = W - S_ + 1;
-  const int xn = C * H * W;
   const int wn = K * C * R * S_;
-  float* xs = smem;        // [C*H*W]
-  float* ws = smem + xn;   // [K*C*R*S]
+  float* ws = smem;  // [K*C*R*S] — shared by all threads, hot across K
+  for (int i = threadIdx.x; i < wn; i += blockDim.x) ws[i] = w[i];
+  __syncthreads();
 
-  for (int b = blockIdx.x; b < B; b += gridDim.x) {
-    // stage input plane + weights
-    for (int i = threadIdx.x; i < xn; i += blockDim.x)
-      xs[i] = x[(int64_t)b * xn + i];
-    for (int i = threadIdx.x; i < wn; i += blockDim.x)
-      ws[i] = w[i];
-    __syncthreads();
-
-    const int on = K * OH * OW;
-    for (int i = threadIdx.x; i < on; i += blockDim.x) {
-      const int k = i / (OH * OW);
-      const int oh = (i / OW) % OH;
-      const int ow = i % OW;
-      float acc = bias ? bias[k] : 0.f;
-      const float* wk = ws + k * C * R * S_;
-      for (int c = 0; c < C; ++c) {
-        const float* xc = xs + c * H * W;
-        const float* wc = wk + c * R * S_;
+  // one thread per output element across the whole tensor: fills the
+  // 256-CU chip at any batch size (the per-batch-block form left half
+  // the chip idle at B=128); x is tiny and L2-resident.
+  const int64_t n_out = (int64_t)B * K * OH * OW;
+  const int xn = C * H * W;
+  const int on = K * OH * OW;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n_out; i += (int64_t)gridDim.x * blockDim.x) {
+    const int b = (int)(i / on);
+    const int k = (int)((i / (OH * OW)) % K);
+    const int oh = (int)((i / OW) % OH);
+    const int ow = (int)(i % OW);
+    float acc = bias ? bias[k] : 0.f;
+    const float* xb = x + (int64_t)b * xn;
+    const float* wk = ws + k * C * R * S_;
+    for (int c = 0; c < C; ++c) {
+      const float* xc = xb + c * H * W + oh * W + ow;
+      const float* wc = wk + c * R * S_;
+      #pragma unroll 5
+      for (int r = 0; r < R; ++r) {
+        const float* xrow = xc + r * W;
+        const float* wrow = wc + r * S_;
+        float a = 0.f;
         #pragma unroll 5
-        for (int r = 0; r < R; ++r) {
-          const float* xrow = xc + (oh + r) * W + ow;
-          const float* wrow = wc + r * S_;
-          float a = 0.f;
-          #pragma unroll 5
-          for (int s = 0; s < S_; ++s) a += xrow[s] * wrow[s];
-          acc += a;
-        }
+        for (int s = 0; s < S_; ++s) a += xrow[s] * wrow[s];
+        acc += a;
       }
-      out[(int64_t)b * on + i] = acc;
     }
-    __syncthreads();
+    out[i] = acc;
   }
 }
 
@@ -111,42 +109,37 @@ __global__ void conv2d_bwd_x_kernel(const float* __restrict__ gy,
                                     int K, int R, int S_) {
   extern __shared__ __attribute__((aligned(16))) float smem[];
   const int OH = H - R + 1, OW = W - S_ + 1;
-  const int gn = K * OH * OW;
   const int wn = K * C * R * S_;
-  float* gys = smem;
-  float* ws = smem + gn;
+  float* ws = smem;
+  for (int i = threadIdx.x; i < wn; i += blockDim.x) ws[i] = w[i];
+  __syncthreads();
 
-  for (int b = blockIdx.x; b < B; b += gridDim.x) {
-    for (int i = threadIdx.x; i < gn; i += blockDim.x)
-      gys[i] = gy[(int64_t)b * gn + i];
-    for (int i = threadIdx.x; i < wn; i += blockDim.x)
-      ws[i] = w[i];
-    __syncthreads();
-
-    const int xn = C * H * W;
-    for (int i = threadIdx.x; i < xn; i += blockDim.x) {
-      const int c = i / (H * W);
-      const int h = (i / W) % H;
-      const int wcol = i % W;
-      float acc = 0.f;
-      for (int k = 0; k < K; ++k) {
-        const float* gk = gys + k * OH * OW;
-        const float* wk = ws + (k * C + c) * R * S_;
-        #pragma unroll 5
-        for (int r = 0; r < R; ++r) {
-          const int oh = h - r;
-          if (oh < 0 || oh >= OH) continue;
-          #pragma unroll 5
-          for (int s = 0; s < S_; ++s) {
-            const int ow = wcol - s;
-            if (ow < 0 || ow >= OW) continue;
-            acc += gk[oh * OW + ow] * wk[r * S_ + s];
-          }
-        }
+  const int64_t n_in = (int64_t)B * C * H * W;
+  const int xn = C * H * W;
+  const int gn = K * OH * OW;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n_in; i += (int64_t)gridDim.x * blockDim.x) {
+    const int b = (int)(i / xn);
+    const int c = (int)((i / (H * W)) % C);
+    const int h = (int)((i / W) % H);
+    const int wcol = (int)(i % W);
+    const float* gb = gy + (int64_t)b * gn;
+    float acc = 0.f;
+    const int r0 = max(0, h - OH + 1), r1 = min(R, h + 1);
+    const int s0 = max(0, wcol - OW + 1), s1 = min(S_, wcol + 1);
+    for (int k = 0; k < K; ++k) {
+      const float* gk = gb + k * OH * OW;
+      const float* wk = ws + (k * C + c) * R * S_;
+      for (int r = r0; r < r1; ++r) {
+        const int oh = h - r;
+        const float* grow = gk + oh * OW + wcol;
+        const float* wrow = wk + r * S_;
+        float a = 0.f;
+        for (int s = s0; s < s1; ++s) a += grow[-s] * wrow[s];
+        acc += a;
       }
-      gx[(int64_t)b * xn + i] = acc;
     }
-    __syncthreads();
+    gx[i] = acc;
   }
 }
 
@@ -157,50 +150,49 @@ __global__ void conv2d_bwd_w_kernel(const float* __restrict__ x,
                                     float* __restrict__ gw,
                                     float* __restrict__ gb,
                                     int B, int C, int H, int W,
-                                    int K, int R, int S_) {
-  extern __shared__ __attribute__((aligned(16))) float smem[];
+                                    int K, int R, int S_, int bchunk) {
   const int OH = H - R + 1, OW = W - S_ + 1;
   const int xn = C * H * W;
   const int gn = K * OH * OW;
   const int wn = K * C * R * S_;
-  float* xs = smem;            // [xn]
-  float* gys = smem + xn;      // [gn]
-  float* wacc = gys + gn;      // [wn]
+  const int nchunks = (B + bchunk - 1) / bchunk;
+  const int chunk = blockIdx.y;
+  const int b0 = chunk * bchunk;
+  const int b1 = min(B, b0 + bchunk);
 
-  for (int b = blockIdx.x; b < B; b += gridDim.x) {
-    for (int i = threadIdx.x; i < xn; i += blockDim.x)
-      xs[i] = x[(int64_t)b * xn + i];
-    for (int i = threadIdx.x; i < gn; i += blockDim.x)
-      gys[i] = gy[(int64_t)b * gn + i];
-    __syncthreads();
-
-    // weight-gradient partial for this batch element
-    for (int i = threadIdx.x; i < wn; i += blockDim.x) {
-      const int k = i / (C * R * S_);
-      const int c = (i / (R * S_)) % C;
-      const int r = (i / S_) % R;
-      const int s = i % S_;
-      const float* gk = gys + k * OH * OW;
-      const float* xc = xs + c * H * W + r * W + s;
-      float acc = 0.f;
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < wn;
+       i += gridDim.x * blockDim.x) {
+    const int k = i / (C * R * S_);
+    const int c = (i / (R * S_)) % C;
+    const int r = (i / S_) % R;
+    const int sx = i % S_;
+    float acc = 0.f;
+    for (int b = b0; b < b1; ++b) {
+      const float* gk = gy + (int64_t)b * gn + k * OH * OW;
+      const float* xc = x + (int64_t)b * xn + c * H * W + r * W + sx;
       for (int oh = 0; oh < OH; ++oh) {
         const float* grow = gk + oh * OW;
         const float* xrow = xc + oh * W;
-        for (int ow = 0; ow < OW; ++ow) acc += grow[ow] * xrow[ow];
-      }
-      atomicAdd(&gw[i], acc);
-    }
-    if (gb) {
-      for (int k = threadIdx.x; k < K; k += blockDim.x) {
-        const float* gk = gys + k * OH * OW;
-        float acc = 0.f;
-        for (int i = 0; i < OH * OW; ++i) acc += gk[i];
-        atomicAdd(&gb[k], acc);
+        float a = 0.f;
+        for (int ow = 0; ow < OW; ++ow) a += grow[ow] * xrow[ow];
+        acc += a;
       }
     }
-    __syncthreads();
+    if (nchunks == 1) gw[i] = acc;
+    else atomicAdd(&gw[i], acc);
   }
-  (void)wacc;
+  if (gb) {
+    for (int k = blockIdx.x * blockDim.x + threadIdx.x; k < K;
+         k += gridDim.x * blockDim.x) {
+      float acc = 0.f;
+      for (int b = b0; b < b1; ++b) {
+        const float* gk = gy + (int64_t)b * gn + k * OH * OW;
+        for (int i2 = 0; i2 < OH * OW; ++i2) acc += gk[i2];
+      }
+      if (nchunks == 1) gb[k] = acc;
+      else atomicAdd(&gb[k], acc);
+    }
+  }
 }
 
 // ===========================================================================
@@ -433,11 +425,13 @@ __global__ void linear_bwd_w_kernel(const float* __restrict__ x,
                                     const float* __restrict__ out,
                                     float* __restrict__ gw,
                                     float* __restrict__ gb,
-                                    int B, int K, int N, int b_per_block) {
-  const int b0 = blockIdx.x * b_per_block;
-  const int b1 = min(B, b0 + b_per_block);
+                                    int B, int K, int N, int bchunk) {
+  const int nchunks = (B + bchunk - 1) / bchunk;
+  const int b0 = blockIdx.y * bchunk;
+  const int b1 = min(B, b0 + bchunk);
   const int wn = N * K;
-  for (int i = threadIdx.x; i < wn; i += blockDim.x) {
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < wn;
+       i += gridDim.x * blockDim.x) {
     const int n = i / K;
     const int k = i % K;
     float acc = 0.f;
@@ -446,17 +440,20 @@ __global__ void linear_bwd_w_kernel(const float* __restrict__ x,
       if (out && out[(int64_t)b * N + n] <= 0.f) g = 0.f;
       acc += g * x[(int64_t)b * K + k];
     }
-    atomicAdd(&gw[i], acc);
+    if (nchunks == 1) gw[i] = acc;
+    else atomicAdd(&gw[i], acc);
   }
   if (gb) {
-    for (int n = threadIdx.x; n < N; n += blockDim.x) {
+    for (int n = blockIdx.x * blockDim.x + threadIdx.x; n < N;
+         n += gridDim.x * blockDim.x) {
       float acc = 0.f;
       for (int b = b0; b < b1; ++b) {
         float g = gy[(int64_t)b * N + n];
         if (out && out[(int64_t)b * N + n] <= 0.f) g = 0.f;
         acc += g;
       }
-      atomicAdd(&gb[n], acc);
+      if (nchunks == 1) gb[n] = acc;
+      else atomicAdd(&gb[n], acc);
     }
   }
 }
@@ -745,6 +742,335 @@ __global__ void scale_f32_kernel(float* __restrict__ dst, float s,
 }
 
 // ===========================================================================
+// Fused whole-Net forward / backward (the launch-bound-regime lever).
+//
+// At the reference's batch sizes the per-op pipeline is bound by ~30
+// kernel launches of ~5 us each, not by arithmetic (rocprof profile,
+// profiles/).  These two kernels run the ENTIRE Net forward
+// (train_dist.py:64-71: conv1 -> pool+relu -> conv2 -> dropout2d ->
+// pool+relu -> fc1+relu -> dropout -> fc2 -> log_softmax -> NLL) and
+// the entire data backward in ONE kernel each, one workgroup per batch
+// element, activations staged through LDS; only the small per-batch
+// weight-gradient reductions stay as separate (chunked) kernels.
+// Architecture constants are Net's (SURVEY.md §2.1 'Net'): they are
+// hard-coded, which is the point — this is the flagship model's fast
+// path; the modular kernels above remain the general path.
+// ===========================================================================
+#define N_C1K 10        // conv1 out channels
+#define N_C2K 20        // conv2 out channels
+#define N_A1 (10*24*24) // conv1 out 5760
+#define N_P1 (10*12*12) // pool1 out 1440
+#define N_A2 (20*8*8)   // conv2 out 1280
+#define N_P2 320        // pool2 out (= fc1 in)
+#define N_H1 50         // fc1 out
+#define N_CLS 10        // classes
+
+__global__ void
+__launch_bounds__(256)
+net_fused_fwd_kernel(
+    const float* __restrict__ x,      // [B,1,28,28]
+    const float* __restrict__ w1, const float* __restrict__ b1,
+    const float* __restrict__ w2, const float* __restrict__ b2,
+    const float* __restrict__ wf1, const float* __restrict__ bf1,
+    const float* __restrict__ wf2, const float* __restrict__ bf2,
+    const int64_t* __restrict__ tgt,
+    float* __restrict__ p1_ws,        // [B,1440]
+    uint8_t* __restrict__ idx1_ws,    // [B,1440]
+    uint8_t* __restrict__ m2_ws,      // [B,20]
+    float* __restrict__ p2_ws,        // [B,320]
+    uint8_t* __restrict__ idx2_ws,    // [B,320]
+    float* __restrict__ h1_ws,        // [B,50]
+    uint8_t* __restrict__ m3_ws,      // [B,50]
+    float* __restrict__ d3_ws,        // [B,50]
+    float* __restrict__ logp_ws,      // [B,10]
+    float* __restrict__ loss,         // scalar (pre-zeroed)
+    const unsigned long long* __restrict__ seed_p,
+    int B, int training) {
+  __shared__ __attribute__((aligned(16))) float xs[784];
+  __shared__ float w1s[N_C1K * 25 + N_C1K];
+  __shared__ float a1[N_A1];
+  __shared__ float p1[N_P1];
+  __shared__ float w2s[N_C2K * 10 * 25 + N_C2K];
+  __shared__ float d2[N_A2];
+  __shared__ float p2[N_P2];
+  __shared__ float d3[N_H1];
+  __shared__ float logits[N_CLS];
+  const int tid = threadIdx.x;
+  const uint64_t seed = seed_p[0];
+
+  for (int b = blockIdx.x; b < B; b += gridDim.x) {
+    // stage input + conv weights
+    for (int i = tid; i < 784; i += 256) xs[i] = x[(int64_t)b * 784 + i];
+    for (int i = tid; i < N_C1K * 25; i += 256) w1s[i] = w1[i];
+    if (tid < N_C1K) w1s[N_C1K * 25 + tid] = b1[tid];
+    for (int i = tid; i < N_C2K * 250; i += 256) w2s[i] = w2[i];
+    if (tid < N_C2K) w2s[N_C2K * 250 + tid] = b2[tid];
+    __syncthreads();
+
+    // conv1: 1->10, k5, 28->24
+    for (int i = tid; i < N_A1; i += 256) {
+      const int k = i / 576, oh = (i / 24) % 24, ow = i % 24;
+      const float* wk = w1s + k * 25;
+      const float* xp = xs + oh * 28 + ow;
+      float acc = w1s[N_C1K * 25 + k];
+      #pragma unroll
+      for (int r = 0; r < 5; ++r) {
+        #pragma unroll
+        for (int s = 0; s < 5; ++s) acc += xp[r * 28 + s] * wk[r * 5 + s];
+      }
+      a1[i] = acc;
+    }
+    __syncthreads();
+
+    // pool1 (2x2) + relu, stash p1 + idx1
+    for (int i = tid; i < N_P1; i += 256) {
+      const int c = i / 144, oh = (i / 12) % 12, ow = i % 12;
+      const float* ap = a1 + c * 576 + oh * 2 * 24 + ow * 2;
+      float v0 = ap[0], v1 = ap[1], v2 = ap[24], v3 = ap[25];
+      int am = 0; float m = v0;
+      if (v1 > m) { m = v1; am = 1; }
+      if (v2 > m) { m = v2; am = 2; }
+      if (v3 > m) { m = v3; am = 3; }
+      const float o = m > 0.f ? m : 0.f;
+      p1[i] = o;
+      p1_ws[(int64_t)b * N_P1 + i] = o;
+      idx1_ws[(int64_t)b * N_P1 + i] = (uint8_t)(m > 0.f ? am : (am | 4));
+    }
+    __syncthreads();
+
+    // conv2: 10->20, k5, 12->8, then channelwise dropout (conv2_drop)
+    for (int i = tid; i < N_A2; i += 256) {
+      const int k = i / 64, oh = (i / 8) % 8, ow = i % 8;
+      const float* wk = w2s + k * 250;
+      float acc = w2s[N_C2K * 250 + k];
+      for (int c = 0; c < 10; ++c) {
+        const float* pp = p1 + c * 144 + oh * 12 + ow;
+        const float* wc = wk + c * 25;
+        #pragma unroll
+        for (int r = 0; r < 5; ++r) {
+          #pragma unroll
+          for (int s = 0; s < 5; ++s)
+            acc += pp[r * 12 + s] * wc[r * 5 + s];
+        }
+      }
+      if (training) {
+        const uint32_t rr = mix32(seed, (uint64_t)b * N_C2K + k);
+        acc = (rr >= 0x80000000u) ? acc * 2.f : 0.f;
+      }
+      d2[i] = acc;
+    }
+    if (training && tid < N_C2K) {
+      const uint32_t rr = mix32(seed, (uint64_t)b * N_C2K + tid);
+      m2_ws[(int64_t)b * N_C2K + tid] = rr >= 0x80000000u;
+    }
+    __syncthreads();
+
+    // pool2 + relu -> p2 (fc1 input), stash
+    for (int i = tid; i < N_P2; i += 256) {
+      const int c = i / 16, oh = (i / 4) % 4, ow = i % 4;
+      const float* dp = d2 + c * 64 + oh * 2 * 8 + ow * 2;
+      float v0 = dp[0], v1 = dp[1], v2 = dp[8], v3 = dp[9];
+      int am = 0; float m = v0;
+      if (v1 > m) { m = v1; am = 1; }
+      if (v2 > m) { m = v2; am = 2; }
+      if (v3 > m) { m = v3; am = 3; }
+      const float o = m > 0.f ? m : 0.f;
+      p2[i] = o;
+      p2_ws[(int64_t)b * N_P2 + i] = o;
+      idx2_ws[(int64_t)b * N_P2 + i] = (uint8_t)(m > 0.f ? am : (am | 4));
+    }
+    __syncthreads();
+
+    // fc1 (320->50) + relu + dropout: 4 threads per output, shfl reduce
+    if (tid < N_H1 * 4) {
+      const int n = tid >> 2, q = tid & 3;
+      const float* wr = wf1 + n * N_P2 + q * 80;
+      const float* pr = p2 + q * 80;
+      float acc = 0.f;
+      #pragma unroll 4
+      for (int k = 0; k < 80; ++k) acc += pr[k] * wr[k];
+      acc += __shfl_down(acc, 1, 4);
+      acc += __shfl_down(acc, 2, 4);
+      if (q == 0) {
+        float h = acc + bf1[n];
+        h = h > 0.f ? h : 0.f;
+        h1_ws[(int64_t)b * N_H1 + n] = h;
+        float dd = h;
+        uint8_t keep = 1;
+        if (training) {
+          const uint32_t rr = mix32(seed ^ 0x5bd1e995u,
+                                    (uint64_t)b * N_H1 + n);
+          keep = rr >= 0x80000000u;
+          dd = keep ? h * 2.f : 0.f;
+        }
+        m3_ws[(int64_t)b * N_H1 + n] = keep;
+        d3_ws[(int64_t)b * N_H1 + n] = dd;
+        d3[n] = dd;
+      }
+    }
+    __syncthreads();
+
+    // fc2 (50->10)
+    if (tid < N_CLS) {
+      const float* wr = wf2 + tid * N_H1;
+      float acc = bf2[tid];
+      #pragma unroll 10
+      for (int k = 0; k < N_H1; ++k) acc += d3[k] * wr[k];
+      logits[tid] = acc;
+    }
+    __syncthreads();
+
+    // log_softmax + NLL (one lane)
+    if (tid == 0) {
+      float m = logits[0];
+      #pragma unroll
+      for (int i = 1; i < N_CLS; ++i) m = fmaxf(m, logits[i]);
+      float ssum = 0.f;
+      #pragma unroll
+      for (int i = 0; i < N_CLS; ++i) ssum += __expf(logits[i] - m);
+      const float lse = m + __logf(ssum);
+      float lp_t = 0.f;
+      const int64_t t = tgt[b];
+      #pragma unroll
+      for (int i = 0; i < N_CLS; ++i) {
+        const float lp = logits[i] - lse;
+        logp_ws[(int64_t)b * N_CLS + i] = lp;
+        if (i == (int)t) lp_t = lp;
+      }
+      atomicAdd(loss, -lp_t / B);
+    }
+    __syncthreads();
+  }
+}
+
+// Fused data-gradient backward: from loss grad to g_a1 (conv1 output
+// grad) in one kernel; weight gradients are reduced afterwards by the
+// chunked conv2d_bwd_w / linear_bwd_w kernels over the stashes.
+__global__ void
+__launch_bounds__(256)
+net_fused_bwd_kernel(
+    const float* __restrict__ w2, const float* __restrict__ wf1,
+    const float* __restrict__ wf2,
+    const int64_t* __restrict__ tgt,
+    const float* __restrict__ gl,      // dLoss (device scalar)
+    const uint8_t* __restrict__ idx1_ws,
+    const uint8_t* __restrict__ m2_ws,
+    const uint8_t* __restrict__ idx2_ws,
+    const float* __restrict__ h1_ws,
+    const uint8_t* __restrict__ m3_ws,
+    const float* __restrict__ logp_ws,
+    float* __restrict__ glog_ws,       // [B,10]  (fc2 out grad)
+    float* __restrict__ gh1_ws,        // [B,50]  (fc1 pre-relu grad)
+    float* __restrict__ ga2_ws,        // [B,1280] (conv2 out grad)
+    float* __restrict__ ga1_ws,        // [B,5760] (conv1 out grad)
+    int B, int training) {
+  __shared__ __attribute__((aligned(16))) float w2s[N_C2K * 250];
+  __shared__ float glg[N_CLS];
+  __shared__ float gd3[N_H1];
+  __shared__ float gh1[N_H1];
+  __shared__ float gp2[N_P2];
+  __shared__ float gd2[N_A2];
+  const int tid = threadIdx.x;
+
+  for (int i = tid; i < N_C2K * 250; i += 256) w2s[i] = w2[i];
+
+  for (int b = blockIdx.x; b < B; b += gridDim.x) {
+    const float sc = gl[0] / B;
+    // g_logits = (exp(logp) - onehot) * gl / B
+    __syncthreads();
+    if (tid < N_CLS) {
+      const float lp = logp_ws[(int64_t)b * N_CLS + tid];
+      const float g = (__expf(lp) -
+                       (tid == (int)tgt[b] ? 1.f : 0.f)) * sc;
+      glg[tid] = g;
+      glog_ws[(int64_t)b * N_CLS + tid] = g;
+    }
+    __syncthreads();
+
+    // g_d3 = wf2^T g_logits ; through dropout + relu -> g_h1pre
+    if (tid < N_H1) {
+      float acc = 0.f;
+      #pragma unroll
+      for (int n = 0; n < N_CLS; ++n)
+        acc += wf2[n * N_H1 + tid] * glg[n];
+      gd3[tid] = acc;
+      float g = acc;
+      if (training) {
+        g = m3_ws[(int64_t)b * N_H1 + tid] ? g * 2.f : 0.f;
+      }
+      if (h1_ws[(int64_t)b * N_H1 + tid] <= 0.f) g = 0.f;
+      gh1[tid] = g;
+      gh1_ws[(int64_t)b * N_H1 + tid] = g;
+    }
+    __syncthreads();
+
+    // g_p2 = wf1^T g_h1pre  (320 outputs x 50)
+    for (int i = tid; i < N_P2; i += 256) {
+      float acc = 0.f;
+      for (int n = 0; n < N_H1; ++n)
+        acc += wf1[n * N_P2 + i] * gh1[n];
+      gp2[i] = acc;
+    }
+    __syncthreads();
+
+    // pool2 bwd + dropout2d bwd -> g_a2 (conv2 out grad)
+    for (int i = tid; i < N_A2; i += 256) gd2[i] = 0.f;
+    __syncthreads();
+    for (int i = tid; i < N_P2; i += 256) {
+      const int c = i / 16, oh = (i / 4) % 4, ow = i % 4;
+      const uint8_t v = idx2_ws[(int64_t)b * N_P2 + i];
+      if (!(v & 4)) {
+        const int am = v & 3;
+        const int base = c * 64 + oh * 2 * 8 + ow * 2;
+        const int off = (am & 1) + (am >> 1) * 8;
+        gd2[base + off] = gp2[i];
+      }
+    }
+    __syncthreads();
+    for (int i = tid; i < N_A2; i += 256) {
+      const int k = i / 64;
+      float g = gd2[i];
+      if (training)
+        g = m2_ws[(int64_t)b * N_C2K + k] ? g * 2.f : 0.f;
+      gd2[i] = g;
+      ga2_ws[(int64_t)b * N_A2 + i] = g;
+    }
+    __syncthreads();
+
+    // conv2 bwd_x: g_p1 = sum_k g_a2 (*) w2 ; then pool1 bwd -> g_a1
+    // write g_a1 sparsely: zero first, then scatter through idx1
+    for (int i = tid; i < N_A1; i += 256)
+      ga1_ws[(int64_t)b * N_A1 + i] = 0.f;
+    __syncthreads();
+    for (int i = tid; i < N_P1; i += 256) {
+      const int c = i / 144, h = (i / 12) % 12, wc = i % 12;
+      float acc = 0.f;
+      const int r0 = max(0, h - 7), r1 = min(5, h + 1);
+      const int s0 = max(0, wc - 7), s1 = min(5, wc + 1);
+      for (int k = 0; k < N_C2K; ++k) {
+        const float* gk = gd2 + k * 64;
+        const float* wk = w2s + (k * 10 + c) * 25;
+        for (int r = r0; r < r1; ++r) {
+          const float* grow = gk + (h - r) * 8 + wc;
+          const float* wrow = wk + r * 5;
+          for (int s = s0; s < s1; ++s) acc += grow[-s] * wrow[s];
+        }
+      }
+      // pool1 backward: route to argmax unless relu-clipped
+      const uint8_t v = idx1_ws[(int64_t)b * N_P1 + i];
+      if (!(v & 4)) {
+        const int am = v & 3;
+        const int oh = (i / 12) % 12, ow = i % 12;
+        const int base = c * 576 + oh * 2 * 24 + ow * 2;
+        const int off = (am & 1) + (am >> 1) * 24;
+        ga1_ws[(int64_t)b * N_A1 + base + off] = acc;
+      }
+    }
+    __syncthreads();
+  }
+}
+
+// ===========================================================================
 // host launchers + pybind
 // ===========================================================================
 namespace {
@@ -754,8 +1080,10 @@ constexpr int BLK = 256;
 void conv2d_fwd(uintptr_t x, uintptr_t w, uintptr_t bias, uintptr_t out,
                 int B, int C, int H, int W, int K, int R, int S_,
                 uintptr_t stream) {
-  const int lds = (C * H * W + K * C * R * S_) * sizeof(float);
-  hipLaunchKernelGGL(conv2d_fwd_kernel, dim3(grid_for(B, 1)), dim3(BLK),
+  const int lds = K * C * R * S_ * sizeof(float);
+  const int OH = H - R + 1, OW = W - S_ + 1;
+  const int64_t n = (int64_t)B * K * OH * OW;
+  hipLaunchKernelGGL(conv2d_fwd_kernel, dim3(grid_for(n, BLK)), dim3(BLK),
                      lds, S(stream), (const float*)x, (const float*)w,
                      (const float*)bias, (float*)out, B, C, H, W, K, R, S_);
 }
@@ -765,21 +1093,31 @@ void conv2d_bwd(uintptr_t x, uintptr_t w, uintptr_t gy, uintptr_t gx,
                 int K, int R, int S_, uintptr_t stream) {
   const int OH = H - R + 1, OW = W - S_ + 1;
   {
-    const int lds = (K * OH * OW + K * C * R * S_) * sizeof(float);
-    hipLaunchKernelGGL(conv2d_bwd_x_kernel, dim3(grid_for(B, 1)),
+    const int lds = K * C * R * S_ * sizeof(float);
+    const int64_t n = (int64_t)B * C * H * W;
+    hipLaunchKernelGGL(conv2d_bwd_x_kernel, dim3(grid_for(n, BLK)),
                        dim3(BLK), lds, S(stream), (const float*)gy,
                        (const float*)w, (float*)gx, B, C, H, W, K, R, S_);
   }
   {
-    // gw is zeroed by the caller; gb must be zeroed here
-    if (gb)
-      HIP_CHECK(hipMemsetAsync((void*)gb, 0, K * sizeof(float), S(stream)));
-    const int lds =
-        (C * H * W + K * OH * OW) * sizeof(float);
-    hipLaunchKernelGGL(conv2d_bwd_w_kernel, dim3(grid_for(B, 1)),
-                       dim3(BLK), lds, S(stream), (const float*)x,
+    const int wn = K * C * R * S_;
+    // pick the batch chunk so the grid lands near ~512 blocks
+    int bchunk = B;
+    const int wtiles = (wn + BLK - 1) / BLK;
+    while (bchunk > 1 && wtiles * ((B + bchunk - 1) / bchunk) < 512)
+      bchunk = (bchunk + 1) / 2;
+    const int nchunks = (B + bchunk - 1) / bchunk;
+    if (nchunks > 1) {
+      HIP_CHECK(hipMemsetAsync((void*)gw, 0, (size_t)wn * sizeof(float),
+                               S(stream)));
+      if (gb)
+        HIP_CHECK(hipMemsetAsync((void*)gb, 0, K * sizeof(float),
+                                 S(stream)));
+    }
+    hipLaunchKernelGGL(conv2d_bwd_w_kernel, dim3(wtiles, nchunks),
+                       dim3(BLK), 0, S(stream), (const float*)x,
                        (const float*)gy, (float*)gw, (float*)gb, B, C, H,
-                       W, K, R, S_);
+                       W, K, R, S_, bchunk);
   }
 }
 
@@ -874,16 +1212,23 @@ void linear_bwd(uintptr_t x, uintptr_t w, uintptr_t gy, uintptr_t out,
                        (const float*)w, (float*)gx, B, K, N);
   }
   {
-    HIP_CHECK(hipMemsetAsync((void*)gw, 0, (size_t)N * K * sizeof(float),
-                             S(stream)));
-    if (gb)
-      HIP_CHECK(hipMemsetAsync((void*)gb, 0, N * sizeof(float), S(stream)));
-    const int b_per_block = 16;
-    const int blocks = (B + b_per_block - 1) / b_per_block;
-    hipLaunchKernelGGL(linear_bwd_w_kernel, dim3(blocks), dim3(BLK), 0,
-                       S(stream), (const float*)x, (const float*)gy,
-                       (const float*)out, (float*)gw, (float*)gb, B, K, N,
-                       b_per_block);
+    const int wn = N * K;
+    const int wtiles = (wn + BLK - 1) / BLK;
+    int bchunk = B;
+    while (bchunk > 1 && wtiles * ((B + bchunk - 1) / bchunk) < 512)
+      bchunk = (bchunk + 1) / 2;
+    const int nchunks = (B + bchunk - 1) / bchunk;
+    if (nchunks > 1) {
+      HIP_CHECK(hipMemsetAsync((void*)gw, 0, (size_t)wn * sizeof(float),
+                               S(stream)));
+      if (gb)
+        HIP_CHECK(hipMemsetAsync((void*)gb, 0, N * sizeof(float),
+                                 S(stream)));
+    }
+    hipLaunchKernelGGL(linear_bwd_w_kernel, dim3(wtiles, nchunks),
+                       dim3(BLK), 0, S(stream), (const float*)x,
+                       (const float*)gy, (const float*)out, (float*)gw,
+                       (float*)gb, B, K, N, bchunk);
   }
 }
 
@@ -959,6 +1304,99 @@ void sgd_step(const std::vector<uintptr_t>& ps,
   }
 }
 
+void net_fused_fwd(uintptr_t x, uintptr_t w1, uintptr_t b1, uintptr_t w2,
+                   uintptr_t b2, uintptr_t wf1, uintptr_t bf1,
+                   uintptr_t wf2, uintptr_t bf2, uintptr_t tgt,
+                   uintptr_t p1_ws, uintptr_t idx1_ws, uintptr_t m2_ws,
+                   uintptr_t p2_ws, uintptr_t idx2_ws, uintptr_t h1_ws,
+                   uintptr_t m3_ws, uintptr_t d3_ws, uintptr_t logp_ws,
+                   uintptr_t loss, uintptr_t seed_dev, int B,
+                   bool training, uintptr_t stream) {
+  HIP_CHECK(hipMemsetAsync((void*)loss, 0, sizeof(float), S(stream)));
+  if (training)
+    hipLaunchKernelGGL(bump_seed_kernel, dim3(1), dim3(64), 0, S(stream),
+                       (unsigned long long*)seed_dev);
+  hipLaunchKernelGGL(net_fused_fwd_kernel, dim3(grid_for(B, 1)), dim3(256),
+                     0, S(stream), (const float*)x, (const float*)w1,
+                     (const float*)b1, (const float*)w2, (const float*)b2,
+                     (const float*)wf1, (const float*)bf1,
+                     (const float*)wf2, (const float*)bf2,
+                     (const int64_t*)tgt, (float*)p1_ws, (uint8_t*)idx1_ws,
+                     (uint8_t*)m2_ws, (float*)p2_ws, (uint8_t*)idx2_ws,
+                     (float*)h1_ws, (uint8_t*)m3_ws, (float*)d3_ws,
+                     (float*)logp_ws, (float*)loss,
+                     (const unsigned long long*)seed_dev, B,
+                     training ? 1 : 0);
+}
+
+static void launch_linear_gw(uintptr_t xp, uintptr_t gyp, uintptr_t gw,
+                             uintptr_t gb, int B, int K, int N,
+                             uintptr_t stream) {
+  const int wn = N * K;
+  const int wtiles = (wn + BLK - 1) / BLK;
+  int bchunk = B;
+  while (bchunk > 1 && wtiles * ((B + bchunk - 1) / bchunk) < 512)
+    bchunk = (bchunk + 1) / 2;
+  const int nchunks = (B + bchunk - 1) / bchunk;
+  if (nchunks > 1) {
+    HIP_CHECK(hipMemsetAsync((void*)gw, 0, (size_t)wn * sizeof(float),
+                             S(stream)));
+    if (gb)
+      HIP_CHECK(hipMemsetAsync((void*)gb, 0, N * sizeof(float), S(stream)));
+  }
+  hipLaunchKernelGGL(linear_bwd_w_kernel, dim3(wtiles, nchunks), dim3(BLK),
+                     0, S(stream), (const float*)xp, (const float*)gyp,
+                     (const float*)nullptr, (float*)gw, (float*)gb, B, K,
+                     N, bchunk);
+}
+
+static void launch_conv_gw(uintptr_t xp, uintptr_t gyp, uintptr_t gw,
+                           uintptr_t gb, int B, int C, int H, int W, int K,
+                           uintptr_t stream) {
+  const int wn = K * C * 25;
+  const int wtiles = (wn + BLK - 1) / BLK;
+  int bchunk = B;
+  while (bchunk > 1 && wtiles * ((B + bchunk - 1) / bchunk) < 512)
+    bchunk = (bchunk + 1) / 2;
+  const int nchunks = (B + bchunk - 1) / bchunk;
+  if (nchunks > 1) {
+    HIP_CHECK(hipMemsetAsync((void*)gw, 0, (size_t)wn * sizeof(float),
+                             S(stream)));
+    if (gb)
+      HIP_CHECK(hipMemsetAsync((void*)gb, 0, K * sizeof(float), S(stream)));
+  }
+  hipLaunchKernelGGL(conv2d_bwd_w_kernel, dim3(wtiles, nchunks), dim3(BLK),
+                     0, S(stream), (const float*)xp, (const float*)gyp,
+                     (float*)gw, (float*)gb, B, C, H, W, K, 5, 5, bchunk);
+}
+
+void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
+                   uintptr_t tgt, uintptr_t gl,
+                   uintptr_t p1_ws, uintptr_t idx1_ws, uintptr_t m2_ws,
+                   uintptr_t p2_ws, uintptr_t idx2_ws, uintptr_t h1_ws,
+                   uintptr_t m3_ws, uintptr_t d3_ws, uintptr_t logp_ws,
+                   uintptr_t glog_ws, uintptr_t gh1_ws, uintptr_t ga2_ws,
+                   uintptr_t ga1_ws,
+                   uintptr_t gw1, uintptr_t gb1, uintptr_t gw2,
+                   uintptr_t gb2, uintptr_t gwf1, uintptr_t gbf1,
+                   uintptr_t gwf2, uintptr_t gbf2, int B, bool training,
+                   uintptr_t stream) {
+  hipLaunchKernelGGL(net_fused_bwd_kernel, dim3(grid_for(B, 1)), dim3(256),
+                     0, S(stream), (const float*)w2, (const float*)wf1,
+                     (const float*)wf2, (const int64_t*)tgt,
+                     (const float*)gl, (const uint8_t*)idx1_ws,
+                     (const uint8_t*)m2_ws, (const uint8_t*)idx2_ws,
+                     (const float*)h1_ws, (const uint8_t*)m3_ws,
+                     (const float*)logp_ws, (float*)glog_ws,
+                     (float*)gh1_ws, (float*)ga2_ws, (float*)ga1_ws, B,
+                     training ? 1 : 0);
+  // weight-gradient batch reductions over the stashes
+  launch_conv_gw(x, ga1_ws, gw1, gb1, B, 1, 28, 28, N_C1K, stream);
+  launch_conv_gw(p1_ws, ga2_ws, gw2, gb2, B, 10, 12, 12, N_C2K, stream);
+  launch_linear_gw(p2_ws, gh1_ws, gwf1, gbf1, B, N_P2, N_H1, stream);
+  launch_linear_gw(d3_ws, glog_ws, gwf2, gbf2, B, N_H1, N_CLS, stream);
+}
+
 void add_inplace(uintptr_t dst, uintptr_t src, int64_t n, int dtype,
                  uintptr_t stream) {
   if (dtype == 7) {  // ncclFloat32 numbering (dist wrapper's _DTYPE)
@@ -1019,6 +1457,8 @@ PYBIND11_MODULE(_kernels, m) {
   m.def("log_softmax_nll_fwd", &log_softmax_nll_fwd);
   m.def("log_softmax_nll_bwd", &log_softmax_nll_bwd);
   m.def("sgd_step", &sgd_step);
+  m.def("net_fused_fwd", &net_fused_fwd);
+  m.def("net_fused_bwd", &net_fused_bwd);
   m.def("add_inplace", &add_inplace);
   m.def("reduce_columns", &reduce_columns);
   m.def("scale_f32", &scale_f32);

@@ -91,10 +91,12 @@ class _RcclBackend:
             raise RuntimeError(
                 "rccl backend requires a GPU (torch.cuda.is_available() is "
                 "False); use backend='gloo' on CPU-only machines")
-        dev = device_id if device_id is not None \
-            else rank % torch.cuda.device_count()
+        dev = (device_id if device_id is not None else rank) \
+            % torch.cuda.device_count()
         torch.cuda.set_device(dev)
-        self._store = rx.TcpStore(addr, port, rank, world_size,
+        # MASTER_PORT belongs to the launcher's own rendezvous when run
+        # under torch.distributed.run; the native store binds port+1.
+        self._store = rx.TcpStore(addr, port + 1, rank, world_size,
                                   rank == 0, 300_000)
         if rank == 0:
             uid = rx.get_unique_id()

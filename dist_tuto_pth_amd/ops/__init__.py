@@ -65,7 +65,7 @@ class _Conv2d(torch.autograd.Function):
             B, C, H, W = x.shape
             K, _, R, S = w.shape
             gx = torch.empty_like(x)
-            gw = torch.zeros_like(w)
+            gw = torch.empty_like(w)   # fully written by the kernel
             gb = torch.empty(K, device=x.device, dtype=x.dtype) \
                 if ctx.needs_input_grad[2] else None
             k.conv2d_bwd(x.data_ptr(), w.data_ptr(), gy.data_ptr(),
@@ -116,7 +116,8 @@ class _MaxPool2dRelu(torch.autograd.Function):
             (idx,) = ctx.saved_tensors
             k = _k()
             B, C, H, W = ctx.in_shape
-            gx = torch.zeros(ctx.in_shape, device=gy.device, dtype=gy.dtype)
+            gx = torch.empty(ctx.in_shape, device=gy.device,
+                             dtype=gy.dtype)  # kernel writes all slots
             k.maxpool2d_relu_bwd(gy.data_ptr(), idx.data_ptr(),
                                  gx.data_ptr(), B, C, H, W, _stream())
             return gx

@@ -1,0 +1,143 @@
+"""Fused whole-Net training step (GPU fast path).
+
+One kernel for the entire forward (conv1..log_softmax+NLL,
+train_dist.py:64-71 + :120) and one for the data backward, plus four
+chunked weight-gradient reductions — ~8 launches per step instead of
+~30 (the per-op pipeline is launch-bound at the reference's batch
+sizes; rocprof evidence in profiles/).  Numerically identical to the
+modular path in eval mode; dropout masks use the same device-seed
+stream but a different indexing, so train-mode losses match only in
+distribution.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, Tuple
+
+import torch
+
+from ..utils.native import load_native
+from . import _seed_ptr, _stream
+
+_ws_cache: Dict[Tuple, Dict[str, torch.Tensor]] = {}
+
+
+def _ws(B: int, device) -> Dict[str, torch.Tensor]:
+    key = (B, device.index)
+    w = _ws_cache.get(key)
+    if w is None:
+        f = lambda *shape: torch.empty(*shape, device=device)  # noqa: E731
+        u8 = lambda *shape: torch.empty(*shape, device=device,  # noqa: E731
+                                        dtype=torch.uint8)
+        w = {
+            "p1": f(B, 1440), "idx1": u8(B, 1440), "m2": u8(B, 20),
+            "p2": f(B, 320), "idx2": u8(B, 320), "h1": f(B, 50),
+            "m3": u8(B, 50), "d3": f(B, 50), "logp": f(B, 10),
+            "glog": f(B, 10), "gh1": f(B, 50), "ga2": f(B, 1280),
+            "ga1": f(B, 5760), "loss": f(()),
+        }
+        _ws_cache[key] = w
+    return w
+
+
+def attach_flat_grads(net) -> torch.Tensor:
+    """Point every parameter's ``.grad`` at a slice of ONE flat buffer
+    so the DP gradient average is a single all-reduce with no
+    pack/unpack copies (the xGMI-friendly layout: one large message
+    instead of 8 tiny ones — SURVEY.md §5 'bucketed').  Returns the
+    flat buffer."""
+    params = [p for p in net.parameters()]
+    total = sum(p.numel() for p in params)
+    flat = torch.zeros(total, device=params[0].device)
+    off = 0
+    for p in params:
+        p.grad = flat[off:off + p.numel()].view_as(p)
+        off += p.numel()
+    return flat
+
+
+class _NetFusedLoss(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w1, b1, w2, b2, wf1, bf1, wf2, bf2, tgt, training):
+        k = load_native("_kernels")
+        B = x.shape[0]
+        ws = _ws(B, x.device)
+        k.net_fused_fwd(
+            x.data_ptr(), w1.data_ptr(), b1.data_ptr(), w2.data_ptr(),
+            b2.data_ptr(), wf1.data_ptr(), bf1.data_ptr(), wf2.data_ptr(),
+            bf2.data_ptr(), tgt.data_ptr(),
+            ws["p1"].data_ptr(), ws["idx1"].data_ptr(), ws["m2"].data_ptr(),
+            ws["p2"].data_ptr(), ws["idx2"].data_ptr(), ws["h1"].data_ptr(),
+            ws["m3"].data_ptr(), ws["d3"].data_ptr(), ws["logp"].data_ptr(),
+            ws["loss"].data_ptr(), _seed_ptr(x.device), B, training,
+            _stream())
+        ctx.save_for_backward(x, w2, wf1, wf2, tgt)
+        ctx.meta = (B, training)
+        ctx.shapes = [w1.shape, b1.shape, w2.shape, b2.shape, wf1.shape,
+                      bf1.shape, wf2.shape, bf2.shape]
+        return ws["loss"].clone()
+
+    @staticmethod
+    def backward(ctx, gl):
+        k = load_native("_kernels")
+        x, w2, wf1, wf2, tgt = ctx.saved_tensors
+        B, training = ctx.meta
+        ws = _ws(B, x.device)
+        dev = x.device
+        grads = [torch.empty(s, device=dev) for s in ctx.shapes]
+        gl = gl.contiguous()
+        k.net_fused_bwd(
+            x.data_ptr(), w2.data_ptr(), wf1.data_ptr(), wf2.data_ptr(),
+            tgt.data_ptr(), gl.data_ptr(),
+            ws["p1"].data_ptr(), ws["idx1"].data_ptr(), ws["m2"].data_ptr(),
+            ws["p2"].data_ptr(), ws["idx2"].data_ptr(), ws["h1"].data_ptr(),
+            ws["m3"].data_ptr(), ws["d3"].data_ptr(), ws["logp"].data_ptr(),
+            ws["glog"].data_ptr(), ws["gh1"].data_ptr(),
+            ws["ga2"].data_ptr(), ws["ga1"].data_ptr(),
+            *[g.data_ptr() for g in grads], B, training, _stream())
+        return (None, *grads, None, None)
+
+
+def net_fused_step(net, x: torch.Tensor, tgt: torch.Tensor) -> torch.Tensor:
+    """Forward + backward in one call, OVERWRITING each parameter's
+    ``.grad`` (no autograd graph, no accumulate-add kernels — the
+    minimal-launch training step used by bench.py).  Returns the loss
+    as a device scalar (no host sync)."""
+    k = load_native("_kernels")
+    B = x.shape[0]
+    ws = _ws(B, x.device)
+    params = [net.conv1.weight, net.conv1.bias, net.conv2.weight,
+              net.conv2.bias, net.fc1.weight, net.fc1.bias,
+              net.fc2.weight, net.fc2.bias]
+    for p in params:
+        if p.grad is None:
+            p.grad = torch.empty_like(p)
+    if "one" not in ws:
+        ws["one"] = torch.ones((), device=x.device)
+    s = _stream()
+    k.net_fused_fwd(
+        x.data_ptr(), *[p.data_ptr() for p in params], tgt.data_ptr(),
+        ws["p1"].data_ptr(), ws["idx1"].data_ptr(), ws["m2"].data_ptr(),
+        ws["p2"].data_ptr(), ws["idx2"].data_ptr(), ws["h1"].data_ptr(),
+        ws["m3"].data_ptr(), ws["d3"].data_ptr(), ws["logp"].data_ptr(),
+        ws["loss"].data_ptr(), _seed_ptr(x.device), B, net.training, s)
+    k.net_fused_bwd(
+        x.data_ptr(), params[2].data_ptr(), params[4].data_ptr(),
+        params[6].data_ptr(), tgt.data_ptr(), ws["one"].data_ptr(),
+        ws["p1"].data_ptr(), ws["idx1"].data_ptr(), ws["m2"].data_ptr(),
+        ws["p2"].data_ptr(), ws["idx2"].data_ptr(), ws["h1"].data_ptr(),
+        ws["m3"].data_ptr(), ws["d3"].data_ptr(), ws["logp"].data_ptr(),
+        ws["glog"].data_ptr(), ws["gh1"].data_ptr(), ws["ga2"].data_ptr(),
+        ws["ga1"].data_ptr(),
+        *[p.grad.data_ptr() for p in params], B, net.training, s)
+    return ws["loss"]
+
+
+def net_fused_loss(net, x: torch.Tensor, tgt: torch.Tensor) -> torch.Tensor:
+    """Mean NLL loss of ``net`` (a models.Net) on (x, tgt), computed by
+    the fused kernels.  Gradients flow to the 8 parameters."""
+    assert x.is_cuda, "fused path is the GPU fast path"
+    return _NetFusedLoss.apply(
+        x.contiguous(), net.conv1.weight, net.conv1.bias,
+        net.conv2.weight, net.conv2.bias, net.fc1.weight, net.fc1.bias,
+        net.fc2.weight, net.fc2.bias, tgt.contiguous(), net.training)
