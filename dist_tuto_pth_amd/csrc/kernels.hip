@@ -838,21 +838,30 @@ net_fused_fwd_kernel(
     }
     __syncthreads();
 
-    // conv2: 10->20, k5, 12->8, then channelwise dropout (conv2_drop)
+    // conv2: 10->20, k5, 12->8, then channelwise dropout (conv2_drop).
+    // Two independent accumulator chains (channel parity) keep the
+    // 250-FMA reduction ILP-bound instead of LDS-latency-bound (there
+    // is ~1 wave/SIMD at reference batch sizes, so TLP cannot hide it).
     for (int i = tid; i < N_A2; i += 256) {
       const int k = i / 64, oh = (i / 8) % 8, ow = i % 8;
       const float* wk = w2s + k * 250;
-      float acc = w2s[N_C2K * 250 + k];
-      for (int c = 0; c < 10; ++c) {
-        const float* pp = p1 + c * 144 + oh * 12 + ow;
-        const float* wc = wk + c * 25;
+      float a0 = w2s[N_C2K * 250 + k], a1 = 0.f;
+      #pragma unroll
+      for (int c = 0; c < 10; c += 2) {
+        const float* pp0 = p1 + c * 144 + oh * 12 + ow;
+        const float* wc0 = wk + c * 25;
+        const float* pp1 = pp0 + 144;
+        const float* wc1 = wc0 + 25;
         #pragma unroll
         for (int r = 0; r < 5; ++r) {
           #pragma unroll
-          for (int s = 0; s < 5; ++s)
-            acc += pp[r * 12 + s] * wc[r * 5 + s];
+          for (int s = 0; s < 5; ++s) {
+            a0 += pp0[r * 12 + s] * wc0[r * 5 + s];
+            a1 += pp1[r * 12 + s] * wc1[r * 5 + s];
+          }
         }
       }
+      float acc = a0 + a1;
       if (training) {
         const uint32_t rr = mix32(seed, (uint64_t)b * N_C2K + k);
         acc = (rr >= 0x80000000u) ? acc * 2.f : 0.f;
@@ -970,6 +979,11 @@ net_fused_bwd_kernel(
   __shared__ float gh1[N_H1];
   __shared__ float gp2[N_P2];
   __shared__ float gd2[N_A2];
+  // zero-padded conv2-out grad [k][16][16]: entry (k,oh+4,ow+4) holds
+  // gd2[k][oh][ow]; the pad makes the transposed-conv window reads
+  // branch-free so the 500-FMA loop unrolls with ILP (the branchy form
+  // was latency-bound at 1 wave/SIMD).
+  __shared__ float gd2p[N_C2K * 256];
   const int tid = threadIdx.x;
 
   for (int i = tid; i < N_C2K * 250; i += 256) w2s[i] = w2[i];
@@ -1027,46 +1041,213 @@ net_fused_bwd_kernel(
       }
     }
     __syncthreads();
+    for (int i = tid; i < N_C2K * 256; i += 256) gd2p[i] = 0.f;
+    __syncthreads();
     for (int i = tid; i < N_A2; i += 256) {
       const int k = i / 64;
       float g = gd2[i];
       if (training)
         g = m2_ws[(int64_t)b * N_C2K + k] ? g * 2.f : 0.f;
-      gd2[i] = g;
       ga2_ws[(int64_t)b * N_A2 + i] = g;
+      const int oh = (i / 8) % 8, ow = i % 8;
+      gd2p[k * 256 + (oh + 4) * 16 + (ow + 4)] = g;
     }
     __syncthreads();
 
-    // conv2 bwd_x: g_p1 = sum_k g_a2 (*) w2 ; then pool1 bwd -> g_a1
-    // write g_a1 sparsely: zero first, then scatter through idx1
-    for (int i = tid; i < N_A1; i += 256)
-      ga1_ws[(int64_t)b * N_A1 + i] = 0.f;
-    __syncthreads();
+    // conv2 bwd_x through the padded tile (fixed 5x5 bounds, two
+    // independent accumulator chains) + pool1 bwd: each pooled
+    // position owns its 2x2 window exclusively, so all four slots are
+    // written here — no zeroing pass.
     for (int i = tid; i < N_P1; i += 256) {
       const int c = i / 144, h = (i / 12) % 12, wc = i % 12;
-      float acc = 0.f;
-      const int r0 = max(0, h - 7), r1 = min(5, h + 1);
-      const int s0 = max(0, wc - 7), s1 = min(5, wc + 1);
-      for (int k = 0; k < N_C2K; ++k) {
-        const float* gk = gd2 + k * 64;
-        const float* wk = w2s + (k * 10 + c) * 25;
-        for (int r = r0; r < r1; ++r) {
-          const float* grow = gk + (h - r) * 8 + wc;
-          const float* wrow = wk + r * 5;
-          for (int s = s0; s < s1; ++s) acc += grow[-s] * wrow[s];
+      float a0 = 0.f, a1 = 0.f;
+      for (int k = 0; k < N_C2K; k += 2) {
+        const float* g0 = gd2p + k * 256;
+        const float* g1 = g0 + 256;
+        const float* w0 = w2s + (k * 10 + c) * 25;
+        const float* w1v = w0 + 250;
+        #pragma unroll
+        for (int r = 0; r < 5; ++r) {
+          const float* gr0 = g0 + (h - r + 4) * 16 + (wc + 4);
+          const float* gr1 = g1 + (h - r + 4) * 16 + (wc + 4);
+          #pragma unroll
+          for (int s = 0; s < 5; ++s) {
+            a0 += gr0[-s] * w0[r * 5 + s];
+            a1 += gr1[-s] * w1v[r * 5 + s];
+          }
         }
       }
-      // pool1 backward: route to argmax unless relu-clipped
+      const float acc = a0 + a1;
       const uint8_t v = idx1_ws[(int64_t)b * N_P1 + i];
-      if (!(v & 4)) {
-        const int am = v & 3;
-        const int oh = (i / 12) % 12, ow = i % 12;
-        const int base = c * 576 + oh * 2 * 24 + ow * 2;
-        const int off = (am & 1) + (am >> 1) * 24;
-        ga1_ws[(int64_t)b * N_A1 + base + off] = acc;
-      }
+      const int am = v & 3;
+      const float g = (v & 4) ? 0.f : acc;
+      const int oh = (i / 12) % 12, ow = i % 12;
+      float* gp = ga1_ws + (int64_t)b * N_A1 + c * 576 +
+                  oh * 2 * 24 + ow * 2;
+      gp[0] = am == 0 ? g : 0.f;
+      gp[1] = am == 1 ? g : 0.f;
+      gp[24] = am == 2 ? g : 0.f;
+      gp[25] = am == 3 ? g : 0.f;
     }
     __syncthreads();
+  }
+}
+
+// Per-chunk partial weight gradients for all four layers in ONE launch
+// (replaces 4 kernels + 8 memsets): grid.x walks tile segments
+// [conv2 | fc1 | conv1 | fc2], grid.y is the batch chunk; partials land
+// in part[chunk][21840] laid out exactly like the flat grad buffer
+// (parameter order conv1.w,b conv2.w,b fc1.w,b fc2.w,b).
+#define GW_TOTAL 21840
+#define OFF_W1 0          // 250
+#define OFF_B1 250        // 10
+#define OFF_W2 260        // 5000
+#define OFF_B2 5260       // 20
+#define OFF_WF1 5280      // 16000
+#define OFF_BF1 21280     // 50
+#define OFF_WF2 21330     // 500
+#define OFF_BF2 21830     // 10
+#define T_CONV2 20        // ceil(5020/256)
+#define T_FC1 63          // ceil(16050/256)
+#define T_CONV1 2         // ceil(260/256)
+#define T_FC2 2           // ceil(510/256)
+#define GW_TILES (T_CONV2 + T_FC1 + T_CONV1 + T_FC2)
+
+__global__ void __launch_bounds__(256)
+net_gw_partial_kernel(const float* __restrict__ x,
+                      const float* __restrict__ p1_ws,
+                      const float* __restrict__ p2_ws,
+                      const float* __restrict__ d3_ws,
+                      const float* __restrict__ ga1_ws,
+                      const float* __restrict__ ga2_ws,
+                      const float* __restrict__ gh1_ws,
+                      const float* __restrict__ glog_ws,
+                      float* __restrict__ part,  // [nch][21840]
+                      int B, int bchunk) {
+  const int tid = threadIdx.x;
+  const int b0 = blockIdx.y * bchunk;
+  const int b1 = min(B, b0 + bchunk);
+  float* my = part + (int64_t)blockIdx.y * GW_TOTAL;
+  int tile = blockIdx.x;
+
+  if (tile < T_CONV2) {  // conv2: gw [20][10][5][5] + gb [20]
+    const int i = tile * 256 + tid;
+    if (i < 5020) {
+      float acc = 0.f;
+      if (i < 5000) {
+        const int k = i / 250, c = (i / 25) % 10;
+        const int r = (i / 5) % 5, sx = i % 5;
+        for (int b = b0; b < b1; ++b) {
+          const float* gk = ga2_ws + (int64_t)b * N_A2 + k * 64;
+          const float* xc = p1_ws + (int64_t)b * N_P1 + c * 144 +
+                            r * 12 + sx;
+          #pragma unroll
+          for (int oh = 0; oh < 8; ++oh) {
+            const float* grow = gk + oh * 8;
+            const float* xrow = xc + oh * 12;
+            #pragma unroll
+            for (int ow = 0; ow < 8; ++ow) acc += grow[ow] * xrow[ow];
+          }
+        }
+        my[OFF_W2 + i] = acc;
+      } else {
+        const int k = i - 5000;
+        for (int b = b0; b < b1; ++b) {
+          const float* gk = ga2_ws + (int64_t)b * N_A2 + k * 64;
+          #pragma unroll
+          for (int j = 0; j < 64; ++j) acc += gk[j];
+        }
+        my[OFF_B2 + k] = acc;
+      }
+    }
+    return;
+  }
+  tile -= T_CONV2;
+  if (tile < T_FC1) {  // fc1: gw [50][320] + gb [50]
+    const int i = tile * 256 + tid;
+    if (i < 16050) {
+      float acc = 0.f;
+      if (i < 16000) {
+        const int n = i / N_P2, k = i % N_P2;
+        for (int b = b0; b < b1; ++b)
+          acc += gh1_ws[(int64_t)b * N_H1 + n] *
+                 p2_ws[(int64_t)b * N_P2 + k];
+        my[OFF_WF1 + i] = acc;
+      } else {
+        const int n = i - 16000;
+        for (int b = b0; b < b1; ++b)
+          acc += gh1_ws[(int64_t)b * N_H1 + n];
+        my[OFF_BF1 + n] = acc;
+      }
+    }
+    return;
+  }
+  tile -= T_FC1;
+  if (tile < T_CONV1) {  // conv1: gw [10][1][5][5] + gb [10]
+    const int i = tile * 256 + tid;
+    if (i < 260) {
+      float acc = 0.f;
+      if (i < 250) {
+        const int k = i / 25, r = (i / 5) % 5, sx = i % 5;
+        for (int b = b0; b < b1; ++b) {
+          const float* gk = ga1_ws + (int64_t)b * N_A1 + k * 576;
+          const float* xc = x + (int64_t)b * 784 + r * 28 + sx;
+          for (int oh = 0; oh < 24; ++oh) {
+            const float* grow = gk + oh * 24;
+            const float* xrow = xc + oh * 28;
+            float a = 0.f;
+            #pragma unroll 8
+            for (int ow = 0; ow < 24; ++ow) a += grow[ow] * xrow[ow];
+            acc += a;
+          }
+        }
+        my[OFF_W1 + i] = acc;
+      } else {
+        const int k = i - 250;
+        for (int b = b0; b < b1; ++b) {
+          const float* gk = ga1_ws + (int64_t)b * N_A1 + k * 576;
+          for (int j = 0; j < 576; ++j) acc += gk[j];
+        }
+        my[OFF_B1 + k] = acc;
+      }
+    }
+    return;
+  }
+  tile -= T_CONV1;
+  {  // fc2: gw [10][50] + gb [10]
+    const int i = tile * 256 + tid;
+    if (i < 510) {
+      float acc = 0.f;
+      if (i < 500) {
+        const int n = i / N_H1, k = i % N_H1;
+        for (int b = b0; b < b1; ++b)
+          acc += glog_ws[(int64_t)b * N_CLS + n] *
+                 d3_ws[(int64_t)b * N_H1 + k];
+        my[OFF_WF2 + i] = acc;
+      } else {
+        const int n = i - 500;
+        for (int b = b0; b < b1; ++b)
+          acc += glog_ws[(int64_t)b * N_CLS + n];
+        my[OFF_BF2 + n] = acc;
+      }
+    }
+  }
+}
+
+// combine: grads[i] = sum over chunks of part[c][i], written through
+// the 8 per-parameter pointers (which may alias one flat buffer).
+struct GwPtrs { float* p[8]; };
+__global__ void net_gw_combine_kernel(const float* __restrict__ part,
+                                      GwPtrs g, int nch) {
+  const int off[9] = {OFF_W1, OFF_B1, OFF_W2, OFF_B2, OFF_WF1, OFF_BF1,
+                      OFF_WF2, OFF_BF2, GW_TOTAL};
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < GW_TOTAL;
+       i += gridDim.x * blockDim.x) {
+    float acc = 0.f;
+    for (int c = 0; c < nch; ++c) acc += part[(int64_t)c * GW_TOTAL + i];
+    int t = 0;
+    while (i >= off[t + 1]) ++t;
+    g.p[t][i - off[t]] = acc;
   }
 }
 
@@ -1376,7 +1557,7 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
                    uintptr_t p2_ws, uintptr_t idx2_ws, uintptr_t h1_ws,
                    uintptr_t m3_ws, uintptr_t d3_ws, uintptr_t logp_ws,
                    uintptr_t glog_ws, uintptr_t gh1_ws, uintptr_t ga2_ws,
-                   uintptr_t ga1_ws,
+                   uintptr_t ga1_ws, uintptr_t part_ws,
                    uintptr_t gw1, uintptr_t gb1, uintptr_t gw2,
                    uintptr_t gb2, uintptr_t gwf1, uintptr_t gbf1,
                    uintptr_t gwf2, uintptr_t gbf2, int B, bool training,
@@ -1390,11 +1571,24 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
                      (const float*)logp_ws, (float*)glog_ws,
                      (float*)gh1_ws, (float*)ga2_ws, (float*)ga1_ws, B,
                      training ? 1 : 0);
-  // weight-gradient batch reductions over the stashes
-  launch_conv_gw(x, ga1_ws, gw1, gb1, B, 1, 28, 28, N_C1K, stream);
-  launch_conv_gw(p1_ws, ga2_ws, gw2, gb2, B, 10, 12, 12, N_C2K, stream);
-  launch_linear_gw(p2_ws, gh1_ws, gwf1, gbf1, B, N_P2, N_H1, stream);
-  launch_linear_gw(d3_ws, glog_ws, gwf2, gbf2, B, N_H1, N_CLS, stream);
+  // weight gradients: one segmented partial kernel over <=8 batch
+  // chunks + one combine (no memsets, no atomics)
+  const int bchunk = (B + 7) / 8;
+  const int nch = (B + bchunk - 1) / bchunk;
+  hipLaunchKernelGGL(net_gw_partial_kernel, dim3(GW_TILES, nch),
+                     dim3(256), 0, S(stream), (const float*)x,
+                     (const float*)p1_ws, (const float*)p2_ws,
+                     (const float*)d3_ws, (const float*)ga1_ws,
+                     (const float*)ga2_ws, (const float*)gh1_ws,
+                     (const float*)glog_ws, (float*)part_ws, B, bchunk);
+  GwPtrs gp;
+  gp.p[0] = (float*)gw1; gp.p[1] = (float*)gb1;
+  gp.p[2] = (float*)gw2; gp.p[3] = (float*)gb2;
+  gp.p[4] = (float*)gwf1; gp.p[5] = (float*)gbf1;
+  gp.p[6] = (float*)gwf2; gp.p[7] = (float*)gbf2;
+  hipLaunchKernelGGL(net_gw_combine_kernel,
+                     dim3((GW_TOTAL + 255) / 256), dim3(256), 0,
+                     S(stream), (const float*)part_ws, gp, nch);
 }
 
 void add_inplace(uintptr_t dst, uintptr_t src, int64_t n, int dtype,

@@ -34,7 +34,7 @@ def _ws(B: int, device) -> Dict[str, torch.Tensor]:
             "p2": f(B, 320), "idx2": u8(B, 320), "h1": f(B, 50),
             "m3": u8(B, 50), "d3": f(B, 50), "logp": f(B, 10),
             "glog": f(B, 10), "gh1": f(B, 50), "ga2": f(B, 1280),
-            "ga1": f(B, 5760), "loss": f(()),
+            "ga1": f(B, 5760), "part": f(8, 21840), "loss": f(()),
         }
         _ws_cache[key] = w
     return w
@@ -94,6 +94,7 @@ class _NetFusedLoss(torch.autograd.Function):
             ws["m3"].data_ptr(), ws["d3"].data_ptr(), ws["logp"].data_ptr(),
             ws["glog"].data_ptr(), ws["gh1"].data_ptr(),
             ws["ga2"].data_ptr(), ws["ga1"].data_ptr(),
+            ws["part"].data_ptr(),
             *[g.data_ptr() for g in grads], B, training, _stream())
         return (None, *grads, None, None)
 
@@ -128,7 +129,7 @@ def net_fused_step(net, x: torch.Tensor, tgt: torch.Tensor) -> torch.Tensor:
         ws["p2"].data_ptr(), ws["idx2"].data_ptr(), ws["h1"].data_ptr(),
         ws["m3"].data_ptr(), ws["d3"].data_ptr(), ws["logp"].data_ptr(),
         ws["glog"].data_ptr(), ws["gh1"].data_ptr(), ws["ga2"].data_ptr(),
-        ws["ga1"].data_ptr(),
+        ws["ga1"].data_ptr(), ws["part"].data_ptr(),
         *[p.grad.data_ptr() for p in params], B, net.training, s)
     return ws["loss"]
 

