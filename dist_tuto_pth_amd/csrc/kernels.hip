@@ -1109,7 +1109,7 @@ net_fused_bwd_kernel(
 #define OFF_BF2 21830     // 10
 #define T_CONV2 20        // ceil(5020/256)
 #define T_FC1 63          // ceil(16050/256)
-#define T_CONV1 2         // ceil(260/256)
+#define T_CONV1 1         // single block: item-parallel + LDS reduce
 #define T_FC2 2           // ceil(510/256)
 #define GW_TILES (T_CONV2 + T_FC1 + T_CONV1 + T_FC2)
 
@@ -1184,33 +1184,39 @@ net_gw_partial_kernel(const float* __restrict__ x,
   }
   tile -= T_FC1;
   if (tile < T_CONV1) {  // conv1: gw [10][1][5][5] + gb [10]
-    const int i = tile * 256 + tid;
-    if (i < 260) {
-      float acc = 0.f;
-      if (i < 250) {
-        const int k = i / 25, r = (i / 5) % 5, sx = i % 5;
+    // 250 outputs is too little parallelism for element-per-thread at
+    // this cost (24x24 window x batch): split each output over its 24
+    // output rows -> 6000 independent items, LDS-atomic reduce.
+    __shared__ float wacc[260];
+    for (int i = tid; i < 260; i += 256) wacc[i] = 0.f;
+    __syncthreads();
+    for (int it = tid; it < 6000 + 240; it += 256) {
+      float a = 0.f;
+      if (it < 6000) {
+        const int e = it / 24, oh = it % 24;
+        const int k = e / 25, r = (e / 5) % 5, sx = e % 5;
         for (int b = b0; b < b1; ++b) {
-          const float* gk = ga1_ws + (int64_t)b * N_A1 + k * 576;
-          const float* xc = x + (int64_t)b * 784 + r * 28 + sx;
-          for (int oh = 0; oh < 24; ++oh) {
-            const float* grow = gk + oh * 24;
-            const float* xrow = xc + oh * 28;
-            float a = 0.f;
-            #pragma unroll 8
-            for (int ow = 0; ow < 24; ++ow) a += grow[ow] * xrow[ow];
-            acc += a;
-          }
+          const float* grow = ga1_ws + (int64_t)b * N_A1 + k * 576 +
+                              oh * 24;
+          const float* xrow = x + (int64_t)b * 784 + (oh + r) * 28 + sx;
+          #pragma unroll 8
+          for (int ow = 0; ow < 24; ++ow) a += grow[ow] * xrow[ow];
         }
-        my[OFF_W1 + i] = acc;
+        atomicAdd(&wacc[e], a);
       } else {
-        const int k = i - 250;
+        const int j = it - 6000;
+        const int k = j / 24, oh = j % 24;
         for (int b = b0; b < b1; ++b) {
-          const float* gk = ga1_ws + (int64_t)b * N_A1 + k * 576;
-          for (int j = 0; j < 576; ++j) acc += gk[j];
+          const float* grow = ga1_ws + (int64_t)b * N_A1 + k * 576 +
+                              oh * 24;
+          #pragma unroll 8
+          for (int ow = 0; ow < 24; ++ow) a += grow[ow];
         }
-        my[OFF_B1 + k] = acc;
+        atomicAdd(&wacc[250 + k], a);
       }
     }
+    __syncthreads();
+    for (int i = tid; i < 260; i += 256) my[OFF_W1 + i] = wacc[i];
     return;
   }
   tile -= T_CONV1;
@@ -1571,9 +1577,20 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
                      (const float*)logp_ws, (float*)glog_ws,
                      (float*)gh1_ws, (float*)ga2_ws, (float*)ga1_ws, B,
                      training ? 1 : 0);
-  // weight gradients: one segmented partial kernel over <=8 batch
-  // chunks + one combine (no memsets, no atomics)
-  const int bchunk = (B + 7) / 8;
+  if (B > 512) {
+    // large batch: per-op adaptive chunked reductions amortize their
+    // memsets; the 32-chunk partial scheme would serialize too much
+    // work per thread
+    launch_conv_gw(x, ga1_ws, gw1, gb1, B, 1, 28, 28, N_C1K, stream);
+    launch_conv_gw(p1_ws, ga2_ws, gw2, gb2, B, 10, 12, 12, N_C2K,
+                   stream);
+    launch_linear_gw(p2_ws, gh1_ws, gwf1, gbf1, B, N_P2, N_H1, stream);
+    launch_linear_gw(d3_ws, glog_ws, gwf2, gbf2, B, N_H1, N_CLS, stream);
+    return;
+  }
+  // small batch (the reference regime): one segmented partial kernel
+  // over <=32 batch chunks + one combine (no memsets, no atomics)
+  const int bchunk = (B + 31) / 32;
   const int nch = (B + bchunk - 1) / bchunk;
   hipLaunchKernelGGL(net_gw_partial_kernel, dim3(GW_TILES, nch),
                      dim3(256), 0, S(stream), (const float*)x,
