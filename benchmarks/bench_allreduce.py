@@ -70,6 +70,11 @@ def main():
     args = p.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", 1))
+    if world == 1:
+        print(json.dumps({"metric": "all-reduce bus BW (GB/s)",
+                          "n_gpus": 1, "note": "all-reduce needs >= 2 "
+                          "ranks; launch via torch.distributed.run"}))
+        return
     rank = int(os.environ.get("RANK", 0))
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
     dev_idx = local_rank % torch.cuda.device_count()
