@@ -34,7 +34,15 @@ def main():
     p.add_argument("--no-overlap", action="store_true",
                    help="reference-style average_gradients after "
                         "backward instead of bucketed overlap")
+    p.add_argument("--channels-last", action="store_true",
+                   help="NHWC memory format (MIOpen fast path)")
+    p.add_argument("--no-find", action="store_true",
+                   help="disable MIOpen exhaustive kernel search")
     args = p.parse_args()
+
+    # MIOpen exhaustive find: pick the fastest conv kernel per shape
+    # during warmup (the warmup steps absorb the search cost)
+    torch.backends.cudnn.benchmark = not args.no_find
 
     world = int(os.environ.get("WORLD_SIZE", 1))
     rank = int(os.environ.get("RANK", 0))
@@ -50,6 +58,8 @@ def main():
 
     torch.manual_seed(1234)
     model = resnet50().to(device)
+    if args.channels_last:
+        model = model.to(memory_format=torch.channels_last)
     if world > 1:
         for prm in model.parameters():
             dist.broadcast(prm.data, src=0)
@@ -60,6 +70,8 @@ def main():
 
     g = torch.Generator().manual_seed(1234 + rank)
     x = torch.randn(args.batch, 3, 224, 224, generator=g).to(device)
+    if args.channels_last:
+        x = x.to(memory_format=torch.channels_last)
     tgt = torch.randint(0, 1000, (args.batch,), generator=g).to(device)
     crit = torch.nn.CrossEntropyLoss()
 
@@ -113,6 +125,8 @@ def main():
             "config": {"model": "ResNet-50", "global_batch":
                        args.batch * world, "input": "3x224x224",
                        "parallelism": f"dp{world}",
+                       "channels_last": args.channels_last,
+                       "miopen_find": not args.no_find,
                        "grad_sync": "average_gradients"
                        if args.no_overlap else
                        f"ddp_overlap_{args.bucket_mb}MB"},

@@ -290,6 +290,17 @@ __global__ void bump_seed_kernel(unsigned long long* s) {
     *s += 0x9E3779B97F4A7C15ull;
 }
 
+// fused step prologue: advance the dropout seed (when training) and
+// zero the loss accumulator in ONE tiny launch instead of a bump
+// kernel + a hipMemsetAsync fill kernel (each ~4.5 us of pure launch
+// cost at the reference's batch sizes).
+__global__ void step_prologue_kernel(unsigned long long* s, float* loss) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    if (s) *s += 0x9E3779B97F4A7C15ull;
+    *loss = 0.f;
+  }
+}
+
 __device__ inline uint32_t mix32(uint64_t seed, uint64_t idx) {
   uint64_t z = seed + 0x9E3779B97F4A7C15ull * (idx + 1);
   z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
@@ -972,7 +983,7 @@ net_fused_bwd_kernel(
     float* __restrict__ gh1_ws,        // [B,50]  (fc1 pre-relu grad)
     float* __restrict__ ga2_ws,        // [B,1280] (conv2 out grad)
     float* __restrict__ ga1_ws,        // [B,5760] (conv1 out grad)
-    int B, int training) {
+    int B, int training, int split) {
   __shared__ __attribute__((aligned(16))) float w2s[N_C2K * 250];
   __shared__ float glg[N_CLS];
   __shared__ float gd3[N_H1];
@@ -988,7 +999,13 @@ net_fused_bwd_kernel(
 
   for (int i = tid; i < N_C2K * 250; i += 256) w2s[i] = w2[i];
 
-  for (int b = blockIdx.x; b < B; b += gridDim.x) {
+  // `split` workgroups cooperate on each sample: the cheap early stages
+  // (g_logits .. gd2p staging, ~35k FMA) are recomputed by every
+  // sibling, the hot conv2-bwd_x stage (720k FMA) is partitioned by
+  // pooled position.  At the reference's B=128 this turns 128 resident
+  // workgroups (half the 256 CUs idle) into 256.
+  for (int bb = blockIdx.x; bb < B * split; bb += gridDim.x) {
+    const int b = bb / split, half = bb % split;
     const float sc = gl[0] / B;
     // g_logits = (exp(logp) - onehot) * gl / B
     __syncthreads();
@@ -997,7 +1014,8 @@ net_fused_bwd_kernel(
       const float g = (__expf(lp) -
                        (tid == (int)tgt[b] ? 1.f : 0.f)) * sc;
       glg[tid] = g;
-      glog_ws[(int64_t)b * N_CLS + tid] = g;
+      if (half == 0)
+        glog_ws[(int64_t)b * N_CLS + tid] = g;
     }
     __syncthreads();
 
@@ -1014,7 +1032,8 @@ net_fused_bwd_kernel(
       }
       if (h1_ws[(int64_t)b * N_H1 + tid] <= 0.f) g = 0.f;
       gh1[tid] = g;
-      gh1_ws[(int64_t)b * N_H1 + tid] = g;
+      if (half == 0)
+        gh1_ws[(int64_t)b * N_H1 + tid] = g;
     }
     __syncthreads();
 
@@ -1048,7 +1067,8 @@ net_fused_bwd_kernel(
       float g = gd2[i];
       if (training)
         g = m2_ws[(int64_t)b * N_C2K + k] ? g * 2.f : 0.f;
-      ga2_ws[(int64_t)b * N_A2 + i] = g;
+      if (half == 0)
+        ga2_ws[(int64_t)b * N_A2 + i] = g;
       const int oh = (i / 8) % 8, ow = i % 8;
       gd2p[k * 256 + (oh + 4) * 16 + (ow + 4)] = g;
     }
@@ -1057,8 +1077,11 @@ net_fused_bwd_kernel(
     // conv2 bwd_x through the padded tile (fixed 5x5 bounds, two
     // independent accumulator chains) + pool1 bwd: each pooled
     // position owns its 2x2 window exclusively, so all four slots are
-    // written here — no zeroing pass.
-    for (int i = tid; i < N_P1; i += 256) {
+    // written here — no zeroing pass.  Partitioned across the `split`
+    // sibling workgroups (N_P1=1440 divisible by 1/2/4/8).
+    const int seg = N_P1 / split;
+    const int i_end = (half + 1) * seg;
+    for (int i = half * seg + tid; i < i_end; i += 256) {
       const int c = i / 144, h = (i / 12) % 12, wc = i % 12;
       float a0 = 0.f, a1 = 0.f;
       for (int k = 0; k < N_C2K; k += 2) {
@@ -1109,9 +1132,13 @@ net_fused_bwd_kernel(
 #define OFF_BF2 21830     // 10
 #define T_CONV2 20        // ceil(5020/256)
 #define T_FC1 63          // ceil(16050/256)
-#define T_CONV1 1         // single block: item-parallel + LDS reduce
+#define T_CONV1 4         // 4 sub-blocks over output-row quarters
 #define T_FC2 2           // ceil(510/256)
 #define GW_TILES (T_CONV2 + T_FC1 + T_CONV1 + T_FC2)
+// conv1's 4 sub-blocks write disjoint 260-float slices: sub 0 to the
+// canonical [OFF_W1,OFF_B1] region, subs 1-3 to an extension past
+// GW_TOTAL; the combine kernel folds the extension back in.
+#define GW_ROW (GW_TOTAL + 3 * 260)
 
 __global__ void __launch_bounds__(256)
 net_gw_partial_kernel(const float* __restrict__ x,
@@ -1127,7 +1154,7 @@ net_gw_partial_kernel(const float* __restrict__ x,
   const int tid = threadIdx.x;
   const int b0 = blockIdx.y * bchunk;
   const int b1 = min(B, b0 + bchunk);
-  float* my = part + (int64_t)blockIdx.y * GW_TOTAL;
+  float* my = part + (int64_t)blockIdx.y * GW_ROW;
   int tile = blockIdx.x;
 
   if (tile < T_CONV2) {  // conv2: gw [20][10][5][5] + gb [20]
@@ -1137,18 +1164,25 @@ net_gw_partial_kernel(const float* __restrict__ x,
       if (i < 5000) {
         const int k = i / 250, c = (i / 25) % 10;
         const int r = (i / 5) % 5, sx = i % 5;
+        // two accumulator chains (output-row parity): the 256-FMA
+        // serial chain was latency-bound at ~1 occupancy-hidden wave
+        float ae = 0.f, ao = 0.f;
         for (int b = b0; b < b1; ++b) {
           const float* gk = ga2_ws + (int64_t)b * N_A2 + k * 64;
           const float* xc = p1_ws + (int64_t)b * N_P1 + c * 144 +
                             r * 12 + sx;
           #pragma unroll
-          for (int oh = 0; oh < 8; ++oh) {
+          for (int oh = 0; oh < 8; oh += 2) {
             const float* grow = gk + oh * 8;
             const float* xrow = xc + oh * 12;
             #pragma unroll
-            for (int ow = 0; ow < 8; ++ow) acc += grow[ow] * xrow[ow];
+            for (int ow = 0; ow < 8; ++ow) {
+              ae += grow[ow] * xrow[ow];
+              ao += grow[8 + ow] * xrow[12 + ow];
+            }
           }
         }
+        acc = ae + ao;
         my[OFF_W2 + i] = acc;
       } else {
         const int k = i - 5000;
@@ -1169,9 +1203,18 @@ net_gw_partial_kernel(const float* __restrict__ x,
       float acc = 0.f;
       if (i < 16000) {
         const int n = i / N_P2, k = i % N_P2;
-        for (int b = b0; b < b1; ++b)
-          acc += gh1_ws[(int64_t)b * N_H1 + n] *
-                 p2_ws[(int64_t)b * N_P2 + k];
+        float ae = 0.f, ao = 0.f;
+        int b = b0;
+        for (; b + 1 < b1; b += 2) {
+          ae += gh1_ws[(int64_t)b * N_H1 + n] *
+                p2_ws[(int64_t)b * N_P2 + k];
+          ao += gh1_ws[(int64_t)(b + 1) * N_H1 + n] *
+                p2_ws[(int64_t)(b + 1) * N_P2 + k];
+        }
+        if (b < b1)
+          ae += gh1_ws[(int64_t)b * N_H1 + n] *
+                p2_ws[(int64_t)b * N_P2 + k];
+        acc = ae + ao;
         my[OFF_WF1 + i] = acc;
       } else {
         const int n = i - 16000;
@@ -1186,26 +1229,36 @@ net_gw_partial_kernel(const float* __restrict__ x,
   if (tile < T_CONV1) {  // conv1: gw [10][1][5][5] + gb [10]
     // 250 outputs is too little parallelism for element-per-thread at
     // this cost (24x24 window x batch): split each output over its 24
-    // output rows -> 6000 independent items, LDS-atomic reduce.
+    // output rows -> 6000 independent items, LDS-atomic reduce, and
+    // over 4 sub-blocks (6 output rows each) so the grid column's
+    // straggler block shrinks 4x.  Sub 0 writes the canonical region,
+    // subs 1-3 the extension rows summed by the combine kernel.
+    const int sub = tile;
+    const int oh0 = sub * 6;
     __shared__ float wacc[260];
     for (int i = tid; i < 260; i += 256) wacc[i] = 0.f;
     __syncthreads();
-    for (int it = tid; it < 6000 + 240; it += 256) {
+    for (int it = tid; it < 1500 + 60; it += 256) {
       float a = 0.f;
-      if (it < 6000) {
-        const int e = it / 24, oh = it % 24;
+      if (it < 1500) {
+        const int e = it / 6, oh = oh0 + it % 6;
         const int k = e / 25, r = (e / 5) % 5, sx = e % 5;
         for (int b = b0; b < b1; ++b) {
           const float* grow = ga1_ws + (int64_t)b * N_A1 + k * 576 +
                               oh * 24;
           const float* xrow = x + (int64_t)b * 784 + (oh + r) * 28 + sx;
-          #pragma unroll 8
-          for (int ow = 0; ow < 24; ++ow) a += grow[ow] * xrow[ow];
+          float ae = 0.f, ao = 0.f;
+          #pragma unroll
+          for (int ow = 0; ow < 24; ow += 2) {
+            ae += grow[ow] * xrow[ow];
+            ao += grow[ow + 1] * xrow[ow + 1];
+          }
+          a += ae + ao;
         }
         atomicAdd(&wacc[e], a);
       } else {
-        const int j = it - 6000;
-        const int k = j / 24, oh = j % 24;
+        const int j = it - 1500;
+        const int k = j / 6, oh = oh0 + j % 6;
         for (int b = b0; b < b1; ++b) {
           const float* grow = ga1_ws + (int64_t)b * N_A1 + k * 576 +
                               oh * 24;
@@ -1216,7 +1269,8 @@ net_gw_partial_kernel(const float* __restrict__ x,
       }
     }
     __syncthreads();
-    for (int i = tid; i < 260; i += 256) my[OFF_W1 + i] = wacc[i];
+    float* dst = (sub == 0) ? (my + OFF_W1) : (my + GW_TOTAL + (sub - 1) * 260);
+    for (int i = tid; i < 260; i += 256) dst[i] = wacc[i];
     return;
   }
   tile -= T_CONV1;
@@ -1250,7 +1304,13 @@ __global__ void net_gw_combine_kernel(const float* __restrict__ part,
   for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < GW_TOTAL;
        i += gridDim.x * blockDim.x) {
     float acc = 0.f;
-    for (int c = 0; c < nch; ++c) acc += part[(int64_t)c * GW_TOTAL + i];
+    for (int c = 0; c < nch; ++c) acc += part[(int64_t)c * GW_ROW + i];
+    if (i < 260) {  // conv1 sub-block extension rows (see GW_ROW)
+      for (int c = 0; c < nch; ++c) {
+        const float* ext = part + (int64_t)c * GW_ROW + GW_TOTAL;
+        acc += ext[i] + ext[260 + i] + ext[520 + i];
+      }
+    }
     int t = 0;
     while (i >= off[t + 1]) ++t;
     g.p[t][i - off[t]] = acc;
@@ -1499,10 +1559,9 @@ void net_fused_fwd(uintptr_t x, uintptr_t w1, uintptr_t b1, uintptr_t w2,
                    uintptr_t m3_ws, uintptr_t d3_ws, uintptr_t logp_ws,
                    uintptr_t loss, uintptr_t seed_dev, int B,
                    bool training, uintptr_t stream) {
-  HIP_CHECK(hipMemsetAsync((void*)loss, 0, sizeof(float), S(stream)));
-  if (training)
-    hipLaunchKernelGGL(bump_seed_kernel, dim3(1), dim3(64), 0, S(stream),
-                       (unsigned long long*)seed_dev);
+  hipLaunchKernelGGL(step_prologue_kernel, dim3(1), dim3(64), 0, S(stream),
+                     training ? (unsigned long long*)seed_dev : nullptr,
+                     (float*)loss);
   hipLaunchKernelGGL(net_fused_fwd_kernel, dim3(grid_for(B, 1)), dim3(256),
                      0, S(stream), (const float*)x, (const float*)w1,
                      (const float*)b1, (const float*)w2, (const float*)b2,
@@ -1568,7 +1627,11 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
                    uintptr_t gb2, uintptr_t gwf1, uintptr_t gbf1,
                    uintptr_t gwf2, uintptr_t gbf2, int B, bool training,
                    uintptr_t stream) {
-  hipLaunchKernelGGL(net_fused_bwd_kernel, dim3(grid_for(B, 1)), dim3(256),
+  // enough sibling workgroups per sample to fill the 256 CUs
+  int split = 1;
+  while (split < 8 && B * split < 256) split *= 2;
+  const int nblk = grid_for((int64_t)B * split, 1);
+  hipLaunchKernelGGL(net_fused_bwd_kernel, dim3(nblk), dim3(256),
                      0, S(stream), (const float*)w2, (const float*)wf1,
                      (const float*)wf2, (const int64_t*)tgt,
                      (const float*)gl, (const uint8_t*)idx1_ws,
@@ -1576,7 +1639,7 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
                      (const float*)h1_ws, (const uint8_t*)m3_ws,
                      (const float*)logp_ws, (float*)glog_ws,
                      (float*)gh1_ws, (float*)ga2_ws, (float*)ga1_ws, B,
-                     training ? 1 : 0);
+                     training ? 1 : 0, split);
   if (B > 512) {
     // large batch: per-op adaptive chunked reductions amortize their
     // memsets; the 32-chunk partial scheme would serialize too much

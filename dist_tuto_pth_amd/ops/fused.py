@@ -34,7 +34,9 @@ def _ws(B: int, device) -> Dict[str, torch.Tensor]:
             "p2": f(B, 320), "idx2": u8(B, 320), "h1": f(B, 50),
             "m3": u8(B, 50), "d3": f(B, 50), "logp": f(B, 10),
             "glog": f(B, 10), "gh1": f(B, 50), "ga2": f(B, 1280),
-            "ga1": f(B, 5760), "part": f(32, 21840), "loss": f(()),
+            # partial rows are GW_ROW = 21840 + 3*260 wide (conv1's 4
+            # weight-grad sub-blocks write disjoint slices; kernels.hip)
+            "ga1": f(B, 5760), "part": f(32, 22620), "loss": f(()),
         }
         _ws_cache[key] = w
     return w
