@@ -47,6 +47,11 @@ def parse_args():
     p.add_argument("--no-fused", action="store_true",
                    help="use the modular per-op kernel pipeline instead "
                         "of the fused whole-Net kernels")
+    p.add_argument("--megakernel", action="store_true",
+                   help="run the whole step as ONE cooperative kernel "
+                        "launch (measured slower than the 6-dispatch "
+                        "fused path at B=128: the weight-gradient phase "
+                        "loses workgroup parallelism — profiles/)")
     return p.parse_args()
 
 
@@ -81,13 +86,27 @@ def main():
 
     use_fused = not args.no_fused and args.mode == "average_gradients"
     flat_grads = None
+    use_mega = False
     if use_fused:
         from dist_tuto_pth_amd.ops.fused import (attach_flat_grads,
-                                                 net_fused_step)
+                                                 net_fused_step,
+                                                 net_fused_train_step,
+                                                 net_step_available)
         flat_grads = attach_flat_grads(model)
+        # single-launch cooperative step kernel (whole step = 1 dispatch);
+        # opt-in: measured slower than the multi-kernel fused path
+        use_mega = (args.megakernel and not args.graph and
+                    args.batch <= 512 and net_step_available())
 
     def step():
         if use_fused:
+            if use_mega:
+                if world == 1:
+                    return net_fused_train_step(model, x, tgt, opt)
+                loss = net_fused_train_step(model, x, tgt, do_sgd=False)
+                dist.all_reduce(flat_grads, op=dist.ReduceOp.AVG)
+                opt.step()
+                return loss
             loss = net_fused_step(model, x, tgt)
             if world > 1:
                 # one flat all-reduce with built-in averaging: the
@@ -164,6 +183,7 @@ def main():
                 "parallelism": f"dp{world}",
                 "grad_sync": args.mode,
                 "fused": use_fused,
+                "megakernel": use_mega,
                 "graph": bool(args.graph),
             },
         }))

@@ -15,6 +15,7 @@
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
+#include <hip/hip_cooperative_groups.h>
 
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
@@ -776,40 +777,29 @@ __global__ void scale_f32_kernel(float* __restrict__ dst, float s,
 #define N_H1 50         // fc1 out
 #define N_CLS 10        // classes
 
-__global__ void
-__launch_bounds__(256)
-net_fused_fwd_kernel(
-    const float* __restrict__ x,      // [B,1,28,28]
+// One sample's full forward (conv1..log_softmax+NLL) by one 256-thread
+// workgroup, activations staged through caller-provided LDS buffers.
+// Returns -— on tid 0 only — the sample's log-prob at the target
+// (callers accumulate the NLL loss); other lanes return 0.
+// Shared by net_fused_fwd_kernel and the single-launch net_step_kernel.
+__device__ __forceinline__ float net_fwd_sample(
+    int b, int tid, int B, int training, uint64_t seed,
+    const float* __restrict__ x,
     const float* __restrict__ w1, const float* __restrict__ b1,
     const float* __restrict__ w2, const float* __restrict__ b2,
     const float* __restrict__ wf1, const float* __restrict__ bf1,
     const float* __restrict__ wf2, const float* __restrict__ bf2,
     const int64_t* __restrict__ tgt,
-    float* __restrict__ p1_ws,        // [B,1440]
-    uint8_t* __restrict__ idx1_ws,    // [B,1440]
-    uint8_t* __restrict__ m2_ws,      // [B,20]
-    float* __restrict__ p2_ws,        // [B,320]
-    uint8_t* __restrict__ idx2_ws,    // [B,320]
-    float* __restrict__ h1_ws,        // [B,50]
-    uint8_t* __restrict__ m3_ws,      // [B,50]
-    float* __restrict__ d3_ws,        // [B,50]
-    float* __restrict__ logp_ws,      // [B,10]
-    float* __restrict__ loss,         // scalar (pre-zeroed)
-    const unsigned long long* __restrict__ seed_p,
-    int B, int training) {
-  __shared__ __attribute__((aligned(16))) float xs[784];
-  __shared__ float w1s[N_C1K * 25 + N_C1K];
-  __shared__ float a1[N_A1];
-  __shared__ float p1[N_P1];
-  __shared__ float w2s[N_C2K * 10 * 25 + N_C2K];
-  __shared__ float d2[N_A2];
-  __shared__ float p2[N_P2];
-  __shared__ float d3[N_H1];
-  __shared__ float logits[N_CLS];
-  const int tid = threadIdx.x;
-  const uint64_t seed = seed_p[0];
-
-  for (int b = blockIdx.x; b < B; b += gridDim.x) {
+    float* __restrict__ p1_ws, uint8_t* __restrict__ idx1_ws,
+    uint8_t* __restrict__ m2_ws, float* __restrict__ p2_ws,
+    uint8_t* __restrict__ idx2_ws, float* __restrict__ h1_ws,
+    uint8_t* __restrict__ m3_ws, float* __restrict__ d3_ws,
+    float* __restrict__ logp_ws,
+    // LDS carve (sizes: 784, 260, 5760, 1440, 5020, 1280, 320, 50, 10)
+    float* xs, float* w1s, float* a1, float* p1, float* w2s, float* d2,
+    float* p2, float* d3, float* logits) {
+  float ret = 0.f;
+  {
     // stage input + conv weights
     for (int i = tid; i < 784; i += 256) xs[i] = x[(int64_t)b * 784 + i];
     for (int i = tid; i < N_C1K * 25; i += 256) w1s[i] = w1[i];
@@ -818,18 +808,30 @@ net_fused_fwd_kernel(
     if (tid < N_C2K) w2s[N_C2K * 250 + tid] = b2[tid];
     __syncthreads();
 
-    // conv1: 1->10, k5, 28->24
-    for (int i = tid; i < N_A1; i += 256) {
-      const int k = i / 576, oh = (i / 24) % 24, ow = i % 24;
-      const float* wk = w1s + k * 25;
-      const float* xp = xs + oh * 28 + ow;
-      float acc = w1s[N_C1K * 25 + k];
+    // conv1: 1->10, k5, 28->24.  Two outputs per loop iteration
+    // (channel halves) -> two independent 25-FMA chains in flight
+    // instead of one (the kernel is latency-bound at ~1 wave/SIMD).
+    for (int i = tid; i < N_A1 / 2; i += 256) {
+      const int j = i + N_A1 / 2;
+      const int ka = i / 576, kb = j / 576;
+      const int oha = (i / 24) % 24, owa = i % 24;
+      const int ohb = (j / 24) % 24, owb = j % 24;
+      const float* wka = w1s + ka * 25;
+      const float* wkb = w1s + kb * 25;
+      const float* xpa = xs + oha * 28 + owa;
+      const float* xpb = xs + ohb * 28 + owb;
+      float acca = w1s[N_C1K * 25 + ka];
+      float accb = w1s[N_C1K * 25 + kb];
       #pragma unroll
       for (int r = 0; r < 5; ++r) {
         #pragma unroll
-        for (int s = 0; s < 5; ++s) acc += xp[r * 28 + s] * wk[r * 5 + s];
+        for (int s = 0; s < 5; ++s) {
+          acca += xpa[r * 28 + s] * wka[r * 5 + s];
+          accb += xpb[r * 28 + s] * wkb[r * 5 + s];
+        }
       }
-      a1[i] = acc;
+      a1[i] = acca;
+      a1[j] = accb;
     }
     __syncthreads();
 
@@ -856,7 +858,8 @@ net_fused_fwd_kernel(
     for (int i = tid; i < N_A2; i += 256) {
       const int k = i / 64, oh = (i / 8) % 8, ow = i % 8;
       const float* wk = w2s + k * 250;
-      float a0 = w2s[N_C2K * 250 + k], a1 = 0.f;
+      // four accumulator chains: (channel parity) x (row parity)
+      float a0 = w2s[N_C2K * 250 + k], a1 = 0.f, a2 = 0.f, a3 = 0.f;
       #pragma unroll
       for (int c = 0; c < 10; c += 2) {
         const float* pp0 = p1 + c * 144 + oh * 12 + ow;
@@ -867,12 +870,17 @@ net_fused_fwd_kernel(
         for (int r = 0; r < 5; ++r) {
           #pragma unroll
           for (int s = 0; s < 5; ++s) {
-            a0 += pp0[r * 12 + s] * wc0[r * 5 + s];
-            a1 += pp1[r * 12 + s] * wc1[r * 5 + s];
+            if (r & 1) {
+              a2 += pp0[r * 12 + s] * wc0[r * 5 + s];
+              a3 += pp1[r * 12 + s] * wc1[r * 5 + s];
+            } else {
+              a0 += pp0[r * 12 + s] * wc0[r * 5 + s];
+              a1 += pp1[r * 12 + s] * wc1[r * 5 + s];
+            }
           }
         }
       }
-      float acc = a0 + a1;
+      float acc = (a0 + a1) + (a2 + a3);
       if (training) {
         const uint32_t rr = mix32(seed, (uint64_t)b * N_C2K + k);
         acc = (rr >= 0x80000000u) ? acc * 2.f : 0.f;
@@ -906,9 +914,13 @@ net_fused_fwd_kernel(
       const int n = tid >> 2, q = tid & 3;
       const float* wr = wf1 + n * N_P2 + q * 80;
       const float* pr = p2 + q * 80;
-      float acc = 0.f;
+      float e0 = 0.f, e1 = 0.f;
       #pragma unroll 4
-      for (int k = 0; k < 80; ++k) acc += pr[k] * wr[k];
+      for (int k = 0; k < 80; k += 2) {
+        e0 += pr[k] * wr[k];
+        e1 += pr[k + 1] * wr[k + 1];
+      }
+      float acc = e0 + e1;
       acc += __shfl_down(acc, 1, 4);
       acc += __shfl_down(acc, 2, 4);
       if (q == 0) {
@@ -957,56 +969,78 @@ net_fused_fwd_kernel(
         logp_ws[(int64_t)b * N_CLS + i] = lp;
         if (i == (int)t) lp_t = lp;
       }
-      atomicAdd(loss, -lp_t / B);
+      ret = lp_t;
     }
     __syncthreads();
   }
+  return ret;
 }
 
-// Fused data-gradient backward: from loss grad to g_a1 (conv1 output
-// grad) in one kernel; weight gradients are reduced afterwards by the
-// chunked conv2d_bwd_w / linear_bwd_w kernels over the stashes.
 __global__ void
 __launch_bounds__(256)
-net_fused_bwd_kernel(
-    const float* __restrict__ w2, const float* __restrict__ wf1,
-    const float* __restrict__ wf2,
+net_fused_fwd_kernel(
+    const float* __restrict__ x,      // [B,1,28,28]
+    const float* __restrict__ w1, const float* __restrict__ b1,
+    const float* __restrict__ w2, const float* __restrict__ b2,
+    const float* __restrict__ wf1, const float* __restrict__ bf1,
+    const float* __restrict__ wf2, const float* __restrict__ bf2,
     const int64_t* __restrict__ tgt,
-    const float* __restrict__ gl,      // dLoss (device scalar)
+    float* __restrict__ p1_ws,        // [B,1440]
+    uint8_t* __restrict__ idx1_ws,    // [B,1440]
+    uint8_t* __restrict__ m2_ws,      // [B,20]
+    float* __restrict__ p2_ws,        // [B,320]
+    uint8_t* __restrict__ idx2_ws,    // [B,320]
+    float* __restrict__ h1_ws,        // [B,50]
+    uint8_t* __restrict__ m3_ws,      // [B,50]
+    float* __restrict__ d3_ws,        // [B,50]
+    float* __restrict__ logp_ws,      // [B,10]
+    float* __restrict__ loss,         // scalar (pre-zeroed)
+    const unsigned long long* __restrict__ seed_p,
+    int B, int training) {
+  __shared__ __attribute__((aligned(16))) float xs[784];
+  __shared__ float w1s[N_C1K * 25 + N_C1K];
+  __shared__ float a1[N_A1];
+  __shared__ float p1[N_P1];
+  __shared__ float w2s[N_C2K * 10 * 25 + N_C2K];
+  __shared__ float d2[N_A2];
+  __shared__ float p2[N_P2];
+  __shared__ float d3[N_H1];
+  __shared__ float logits[N_CLS];
+  const int tid = threadIdx.x;
+  const uint64_t seed = seed_p[0];
+
+  for (int b = blockIdx.x; b < B; b += gridDim.x) {
+    const float lp_t = net_fwd_sample(
+        b, tid, B, training, seed, x, w1, b1, w2, b2, wf1, bf1, wf2, bf2,
+        tgt, p1_ws, idx1_ws, m2_ws, p2_ws, idx2_ws, h1_ws, m3_ws, d3_ws,
+        logp_ws, xs, w1s, a1, p1, w2s, d2, p2, d3, logits);
+    if (tid == 0) atomicAdd(loss, -lp_t / B);
+  }
+}
+
+// One (sample, sibling-half)'s data backward: loss grad -> g_a1 (conv1
+// output grad).  `split` sibling workgroups cooperate on each sample:
+// the cheap early stages (g_logits .. gd2p staging, ~35k FMA) are
+// recomputed by every sibling, the hot conv2-bwd_x stage (720k FMA) is
+// partitioned by pooled position.  w2s must be pre-staged with conv2's
+// weights (5000 floats) by the caller.  Shared by net_fused_bwd_kernel
+// and the single-launch net_step_kernel.
+__device__ __forceinline__ void net_bwd_sample(
+    int b, int half, int split, int tid, int B, int training, float sc,
+    const float* __restrict__ wf1, const float* __restrict__ wf2,
+    const int64_t* __restrict__ tgt,
     const uint8_t* __restrict__ idx1_ws,
     const uint8_t* __restrict__ m2_ws,
     const uint8_t* __restrict__ idx2_ws,
     const float* __restrict__ h1_ws,
     const uint8_t* __restrict__ m3_ws,
     const float* __restrict__ logp_ws,
-    float* __restrict__ glog_ws,       // [B,10]  (fc2 out grad)
-    float* __restrict__ gh1_ws,        // [B,50]  (fc1 pre-relu grad)
-    float* __restrict__ ga2_ws,        // [B,1280] (conv2 out grad)
-    float* __restrict__ ga1_ws,        // [B,5760] (conv1 out grad)
-    int B, int training, int split) {
-  __shared__ __attribute__((aligned(16))) float w2s[N_C2K * 250];
-  __shared__ float glg[N_CLS];
-  __shared__ float gd3[N_H1];
-  __shared__ float gh1[N_H1];
-  __shared__ float gp2[N_P2];
-  __shared__ float gd2[N_A2];
-  // zero-padded conv2-out grad [k][16][16]: entry (k,oh+4,ow+4) holds
-  // gd2[k][oh][ow]; the pad makes the transposed-conv window reads
-  // branch-free so the 500-FMA loop unrolls with ILP (the branchy form
-  // was latency-bound at 1 wave/SIMD).
-  __shared__ float gd2p[N_C2K * 256];
-  const int tid = threadIdx.x;
-
-  for (int i = tid; i < N_C2K * 250; i += 256) w2s[i] = w2[i];
-
-  // `split` workgroups cooperate on each sample: the cheap early stages
-  // (g_logits .. gd2p staging, ~35k FMA) are recomputed by every
-  // sibling, the hot conv2-bwd_x stage (720k FMA) is partitioned by
-  // pooled position.  At the reference's B=128 this turns 128 resident
-  // workgroups (half the 256 CUs idle) into 256.
-  for (int bb = blockIdx.x; bb < B * split; bb += gridDim.x) {
-    const int b = bb / split, half = bb % split;
-    const float sc = gl[0] / B;
+    float* __restrict__ glog_ws, float* __restrict__ gh1_ws,
+    float* __restrict__ ga2_ws, float* __restrict__ ga1_ws,
+    // LDS carve (sizes: 5000, 10, 50, 50, 320, 1280, 5120)
+    const float* w2s, float* glg, float* gd3, float* gh1, float* gp2,
+    float* gd2, float* gd2p) {
+  {
     // g_logits = (exp(logp) - onehot) * gl / B
     __syncthreads();
     if (tid < N_CLS) {
@@ -1116,6 +1150,51 @@ net_fused_bwd_kernel(
   }
 }
 
+// Fused data-gradient backward: from loss grad to g_a1 (conv1 output
+// grad) in one kernel; weight gradients are reduced afterwards by the
+// chunked partial/combine kernels over the stashes.
+__global__ void
+__launch_bounds__(256)
+net_fused_bwd_kernel(
+    const float* __restrict__ w2, const float* __restrict__ wf1,
+    const float* __restrict__ wf2,
+    const int64_t* __restrict__ tgt,
+    const float* __restrict__ gl,      // dLoss (device scalar)
+    const uint8_t* __restrict__ idx1_ws,
+    const uint8_t* __restrict__ m2_ws,
+    const uint8_t* __restrict__ idx2_ws,
+    const float* __restrict__ h1_ws,
+    const uint8_t* __restrict__ m3_ws,
+    const float* __restrict__ logp_ws,
+    float* __restrict__ glog_ws,       // [B,10]  (fc2 out grad)
+    float* __restrict__ gh1_ws,        // [B,50]  (fc1 pre-relu grad)
+    float* __restrict__ ga2_ws,        // [B,1280] (conv2 out grad)
+    float* __restrict__ ga1_ws,        // [B,5760] (conv1 out grad)
+    int B, int training, int split) {
+  __shared__ __attribute__((aligned(16))) float w2s[N_C2K * 250];
+  __shared__ float glg[N_CLS];
+  __shared__ float gd3[N_H1];
+  __shared__ float gh1[N_H1];
+  __shared__ float gp2[N_P2];
+  __shared__ float gd2[N_A2];
+  // zero-padded conv2-out grad [k][16][16]: entry (k,oh+4,ow+4) holds
+  // gd2[k][oh][ow]; the pad makes the transposed-conv window reads
+  // branch-free so the 500-FMA loop unrolls with ILP (the branchy form
+  // was latency-bound at 1 wave/SIMD).
+  __shared__ float gd2p[N_C2K * 256];
+  const int tid = threadIdx.x;
+
+  for (int i = tid; i < N_C2K * 250; i += 256) w2s[i] = w2[i];
+  const float sc = gl[0] / B;
+
+  for (int bb = blockIdx.x; bb < B * split; bb += gridDim.x) {
+    net_bwd_sample(bb / split, bb % split, split, tid, B, training, sc,
+                   wf1, wf2, tgt, idx1_ws, m2_ws, idx2_ws, h1_ws, m3_ws,
+                   logp_ws, glog_ws, gh1_ws, ga2_ws, ga1_ws,
+                   w2s, glg, gd3, gh1, gp2, gd2, gd2p);
+  }
+}
+
 // Per-chunk partial weight gradients for all four layers in ONE launch
 // (replaces 4 kernels + 8 memsets): grid.x walks tile segments
 // [conv2 | fc1 | conv1 | fc2], grid.y is the batch chunk; partials land
@@ -1140,23 +1219,20 @@ net_fused_bwd_kernel(
 // GW_TOTAL; the combine kernel folds the extension back in.
 #define GW_ROW (GW_TOTAL + 3 * 260)
 
-__global__ void __launch_bounds__(256)
-net_gw_partial_kernel(const float* __restrict__ x,
-                      const float* __restrict__ p1_ws,
-                      const float* __restrict__ p2_ws,
-                      const float* __restrict__ d3_ws,
-                      const float* __restrict__ ga1_ws,
-                      const float* __restrict__ ga2_ws,
-                      const float* __restrict__ gh1_ws,
-                      const float* __restrict__ glog_ws,
-                      float* __restrict__ part,  // [nch][21840]
-                      int B, int bchunk) {
-  const int tid = threadIdx.x;
-  const int b0 = blockIdx.y * bchunk;
-  const int b1 = min(B, b0 + bchunk);
-  float* my = part + (int64_t)blockIdx.y * GW_ROW;
-  int tile = blockIdx.x;
-
+// One (tile, batch-chunk) partial weight-gradient reduction.  Tiles
+// walk [conv2 | fc1 | conv1x4 | fc2]; partials land in my[GW_ROW]
+// laid out like the flat grad buffer.  Shared by net_gw_partial_kernel
+// and the single-launch net_step_kernel.
+__device__ __forceinline__ void net_gw_tile(
+    int tile, int tid, int b0, int b1, float* __restrict__ my,
+    const float* __restrict__ x,
+    const float* __restrict__ p1_ws,
+    const float* __restrict__ p2_ws,
+    const float* __restrict__ d3_ws,
+    const float* __restrict__ ga1_ws,
+    const float* __restrict__ ga2_ws,
+    const float* __restrict__ gh1_ws,
+    const float* __restrict__ glog_ws) {
   if (tile < T_CONV2) {  // conv2: gw [20][10][5][5] + gb [20]
     const int i = tile * 256 + tid;
     if (i < 5020) {
@@ -1294,26 +1370,204 @@ net_gw_partial_kernel(const float* __restrict__ x,
   }
 }
 
+__global__ void __launch_bounds__(256)
+net_gw_partial_kernel(const float* __restrict__ x,
+                      const float* __restrict__ p1_ws,
+                      const float* __restrict__ p2_ws,
+                      const float* __restrict__ d3_ws,
+                      const float* __restrict__ ga1_ws,
+                      const float* __restrict__ ga2_ws,
+                      const float* __restrict__ gh1_ws,
+                      const float* __restrict__ glog_ws,
+                      float* __restrict__ part,  // [nch][GW_ROW]
+                      int B, int bchunk) {
+  const int b0 = blockIdx.y * bchunk;
+  net_gw_tile(blockIdx.x, threadIdx.x, b0, min(B, b0 + bchunk),
+              part + (int64_t)blockIdx.y * GW_ROW, x, p1_ws, p2_ws,
+              d3_ws, ga1_ws, ga2_ws, gh1_ws, glog_ws);
+}
+
 // combine: grads[i] = sum over chunks of part[c][i], written through
 // the 8 per-parameter pointers (which may alias one flat buffer).
 struct GwPtrs { float* p[8]; };
+// sum flat-grad element i over the nch chunk rows (+ conv1 extension)
+__device__ __forceinline__ float net_gw_combine_elem(
+    int i, int nch, const float* __restrict__ part) {
+  // four independent accumulator chains: the single 32-deep
+  // load+add chain was latency-bound (VALUBusy ~0, profiles/)
+  float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+  int c = 0;
+  for (; c + 3 < nch; c += 4) {
+    a0 += part[(int64_t)c * GW_ROW + i];
+    a1 += part[(int64_t)(c + 1) * GW_ROW + i];
+    a2 += part[(int64_t)(c + 2) * GW_ROW + i];
+    a3 += part[(int64_t)(c + 3) * GW_ROW + i];
+  }
+  for (; c < nch; ++c) a0 += part[(int64_t)c * GW_ROW + i];
+  if (i < 260) {  // conv1 sub-block extension rows (see GW_ROW)
+    for (int c2 = 0; c2 < nch; ++c2) {
+      const float* ext = part + (int64_t)c2 * GW_ROW + GW_TOTAL;
+      a0 += ext[i];
+      a1 += ext[260 + i];
+      a2 += ext[520 + i];
+    }
+  }
+  return (a0 + a1) + (a2 + a3);
+}
+
+__device__ __forceinline__ int net_gw_tensor_of(int i, const int* off) {
+  int t = 0;
+  while (i >= off[t + 1]) ++t;
+  return t;
+}
+
 __global__ void net_gw_combine_kernel(const float* __restrict__ part,
                                       GwPtrs g, int nch) {
   const int off[9] = {OFF_W1, OFF_B1, OFF_W2, OFF_B2, OFF_WF1, OFF_BF1,
                       OFF_WF2, OFF_BF2, GW_TOTAL};
   for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < GW_TOTAL;
        i += gridDim.x * blockDim.x) {
-    float acc = 0.f;
-    for (int c = 0; c < nch; ++c) acc += part[(int64_t)c * GW_ROW + i];
-    if (i < 260) {  // conv1 sub-block extension rows (see GW_ROW)
-      for (int c = 0; c < nch; ++c) {
-        const float* ext = part + (int64_t)c * GW_ROW + GW_TOTAL;
-        acc += ext[i] + ext[260 + i] + ext[520 + i];
-      }
-    }
-    int t = 0;
-    while (i >= off[t + 1]) ++t;
+    const float acc = net_gw_combine_elem(i, nch, part);
+    const int t = net_gw_tensor_of(i, off);
     g.p[t][i - off[t]] = acc;
+  }
+}
+
+// ===========================================================================
+// Single-launch training step (cooperative kernel).
+//
+// The 6-dispatch step (prologue, fwd, bwd, gw-partial, combine, sgd)
+// pays a ~4.5 us device dispatch floor per kernel at the reference's
+// batch sizes (rocprof, profiles/).  This kernel runs the WHOLE step —
+// forward, data backward, weight-gradient reduction, combine and the
+// SGD+momentum update — in ONE hipLaunchCooperativeKernel dispatch,
+// with cooperative-groups grid barriers between phases.  The per-phase
+// work is the same shared __device__ code as the modular kernels
+// (net_fwd_sample / net_bwd_sample / net_gw_tile / net_gw_combine_elem)
+// so numerics are identical by construction.
+//
+// The loss is accumulated per-workgroup (loss_part, no atomics and no
+// pre-zeroing) and summed into loss_out by workgroup 0 in the last
+// phase, which also advances the device dropout seed for the NEXT step
+// (replacing the step-prologue kernel; hipGraph-replay safe).
+//
+// When do_sgd == 0 the kernel stops after writing grads (the DP path:
+// host runs the flat-gradient all-reduce, then sgd_step_kernel).
+// ===========================================================================
+namespace cg = cooperative_groups;
+
+__global__ void __launch_bounds__(256)
+net_step_kernel(
+    const float* __restrict__ x,
+    const int64_t* __restrict__ tgt,
+    const float* __restrict__ gl,
+    float* __restrict__ p1_ws, uint8_t* __restrict__ idx1_ws,
+    uint8_t* __restrict__ m2_ws, float* __restrict__ p2_ws,
+    uint8_t* __restrict__ idx2_ws, float* __restrict__ h1_ws,
+    uint8_t* __restrict__ m3_ws, float* __restrict__ d3_ws,
+    float* __restrict__ logp_ws, float* __restrict__ glog_ws,
+    float* __restrict__ gh1_ws, float* __restrict__ ga2_ws,
+    float* __restrict__ ga1_ws, float* __restrict__ part,
+    float* __restrict__ loss_part,    // [gridDim.x]
+    float* __restrict__ loss_out,     // scalar
+    unsigned long long* __restrict__ seed_p,
+    GwPtrs prm, GwPtrs grd, GwPtrs buf,
+    float lr, float mu, int do_sgd,
+    int B, int training, int split, int bchunk, int nch) {
+  // LDS union: fwd carve (14,924 floats) reused by the bwd carve
+  // (11,830) and the final-phase reduction scratch; phases are
+  // separated by grid barriers.
+  __shared__ __attribute__((aligned(16))) float smem[14924];
+  float* xs = smem;             // 784
+  float* w1s = xs + 784;        // 260
+  float* a1 = w1s + 260;        // 5760
+  float* p1 = a1 + 5760;        // 1440
+  float* w2s = p1 + 1440;       // 5020 (fwd: w2+b2)
+  float* d2 = w2s + 5020;       // 1280
+  float* p2 = d2 + 1280;        // 320
+  float* d3 = p2 + 320;         // 50
+  float* logits = d3 + 50;      // 10
+  float* b_w2s = smem;          // 5000 (bwd: w2 only)
+  float* b_glg = smem + 5008;   // 10
+  float* b_gd3 = smem + 5024;   // 50
+  float* b_gh1 = smem + 5088;   // 50
+  float* b_gp2 = smem + 5152;   // 320
+  float* b_gd2 = smem + 5472;   // 1280
+  float* b_gd2p = smem + 6752;  // 5120
+  const int tid = threadIdx.x;
+  const int wg = blockIdx.x, nblk = gridDim.x;
+  cg::grid_group grid = cg::this_grid();
+
+  const float* w1 = prm.p[0]; const float* b1 = prm.p[1];
+  const float* w2 = prm.p[2]; const float* b2 = prm.p[3];
+  const float* wf1 = prm.p[4]; const float* bf1 = prm.p[5];
+  const float* wf2 = prm.p[6]; const float* bf2 = prm.p[7];
+  const uint64_t seed = seed_p[0];
+
+  // ---- phase 1: forward ------------------------------------------------
+  float lsum = 0.f;
+  for (int b = wg; b < B; b += nblk) {
+    const float lp_t = net_fwd_sample(
+        b, tid, B, training, seed, x, w1, b1, w2, b2, wf1, bf1, wf2, bf2,
+        tgt, p1_ws, idx1_ws, m2_ws, p2_ws, idx2_ws, h1_ws, m3_ws, d3_ws,
+        logp_ws, xs, w1s, a1, p1, w2s, d2, p2, d3, logits);
+    if (tid == 0) lsum += -lp_t / B;
+  }
+  if (tid == 0) loss_part[wg] = lsum;
+  grid.sync();
+
+  // ---- phase 2: data backward -----------------------------------------
+  for (int i = tid; i < N_C2K * 250; i += 256) b_w2s[i] = w2[i];
+  const float sc = gl[0] / B;
+  for (int bb = wg; bb < B * split; bb += nblk) {
+    net_bwd_sample(bb / split, bb % split, split, tid, B, training, sc,
+                   wf1, wf2, tgt, idx1_ws, m2_ws, idx2_ws, h1_ws, m3_ws,
+                   logp_ws, glog_ws, gh1_ws, ga2_ws, ga1_ws,
+                   b_w2s, b_glg, b_gd3, b_gh1, b_gp2, b_gd2, b_gd2p);
+  }
+  grid.sync();
+
+  // ---- phase 3: partial weight gradients ------------------------------
+  for (int t = wg; t < GW_TILES * nch; t += nblk) {
+    const int tile = t % GW_TILES, ch = t / GW_TILES;
+    const int b0 = ch * bchunk;
+    net_gw_tile(tile, tid, b0, min(B, b0 + bchunk),
+                part + (int64_t)ch * GW_ROW, x, p1_ws, p2_ws, d3_ws,
+                ga1_ws, ga2_ws, gh1_ws, glog_ws);
+    __syncthreads();
+  }
+  grid.sync();
+
+  // ---- phase 4: combine (+ SGD), loss finalize, seed bump --------------
+  const int off[9] = {OFF_W1, OFF_B1, OFF_W2, OFF_B2, OFF_WF1, OFF_BF1,
+                      OFF_WF2, OFF_BF2, GW_TOTAL};
+  for (int i = wg * 256 + tid; i < GW_TOTAL; i += nblk * 256) {
+    const float acc = net_gw_combine_elem(i, nch, part);
+    const int t = net_gw_tensor_of(i, off);
+    const int64_t j = i - off[t];
+    grd.p[t][j] = acc;
+    if (do_sgd) {
+      float v = acc;
+      if (buf.p[t]) {
+        v = mu * buf.p[t][j] + acc;
+        buf.p[t][j] = v;
+      }
+      prm.p[t][j] -= lr * v;
+    }
+  }
+  if (wg == 0) {
+    float v = (tid < nblk) ? loss_part[tid] : 0.f;
+    if (tid + 256 < nblk) v += loss_part[tid + 256];  // nblk <= 512
+    smem[tid] = v;
+    __syncthreads();
+    for (int s = 128; s > 0; s >>= 1) {
+      if (tid < s) smem[tid] += smem[tid + s];
+      __syncthreads();
+    }
+    if (tid == 0) {
+      *loss_out = smem[0];
+      if (training) *seed_p = seed + 0x9E3779B97F4A7C15ull;
+    }
   }
 }
 
@@ -1671,6 +1925,98 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
                      S(stream), (const float*)part_ws, gp, nch);
 }
 
+// ---- single-launch training step (cooperative) --------------------------
+int net_step_max_blocks() {
+  static int cached = -2;
+  if (cached != -2) return cached;
+  int dev = 0;
+  if (hipGetDevice(&dev) != hipSuccess) { cached = 0; return 0; }
+  int coop = 0;
+  if (hipDeviceGetAttribute(&coop, hipDeviceAttributeCooperativeLaunch,
+                            dev) != hipSuccess || !coop) {
+    cached = 0;
+    return 0;
+  }
+  int per_cu = 0;
+  if (hipOccupancyMaxActiveBlocksPerMultiprocessor(
+          &per_cu, reinterpret_cast<const void*>(net_step_kernel), 256,
+          0) != hipSuccess || per_cu < 1) {
+    cached = 0;
+    return 0;
+  }
+  hipDeviceProp_t prop;
+  if (hipGetDeviceProperties(&prop, dev) != hipSuccess) {
+    cached = 0;
+    return 0;
+  }
+  int nb = per_cu * prop.multiProcessorCount;
+  if (nb > 512) nb = 512;  // loss_part reduction handles <= 512
+  cached = nb;
+  return nb;
+}
+
+bool net_step_available() { return net_step_max_blocks() > 0; }
+
+void net_step(uintptr_t x, uintptr_t tgt, uintptr_t gl,
+              uintptr_t p1_ws, uintptr_t idx1_ws, uintptr_t m2_ws,
+              uintptr_t p2_ws, uintptr_t idx2_ws, uintptr_t h1_ws,
+              uintptr_t m3_ws, uintptr_t d3_ws, uintptr_t logp_ws,
+              uintptr_t glog_ws, uintptr_t gh1_ws, uintptr_t ga2_ws,
+              uintptr_t ga1_ws, uintptr_t part_ws, uintptr_t loss_part,
+              uintptr_t loss_out, uintptr_t seed_dev,
+              const std::vector<uintptr_t>& prm_v,
+              const std::vector<uintptr_t>& grd_v,
+              const std::vector<uintptr_t>& buf_v,
+              double lr, double mu, bool do_sgd, int B, bool training,
+              uintptr_t stream) {
+  const int nblk = net_step_max_blocks();
+  if (nblk < 1)
+    throw std::runtime_error(
+        "net_step: cooperative launch unavailable on this device");
+  if (prm_v.size() != 8 || grd_v.size() != 8)
+    throw std::runtime_error("net_step: expected 8 param/grad pointers");
+  int split = 1;
+  while (split < 8 && B * split < nblk) split *= 2;
+  int bchunk = (B + 31) / 32;
+  int nch = (B + bchunk - 1) / bchunk;
+  GwPtrs prm{}, grd{}, buf{};
+  for (int i = 0; i < 8; ++i) {
+    prm.p[i] = (float*)prm_v[i];
+    grd.p[i] = (float*)grd_v[i];
+    buf.p[i] = (i < (int)buf_v.size()) ? (float*)buf_v[i] : nullptr;
+  }
+  const float* xp = (const float*)x;
+  const int64_t* tgtp = (const int64_t*)tgt;
+  const float* glp = (const float*)gl;
+  float* p1p = (float*)p1_ws;
+  uint8_t* idx1p = (uint8_t*)idx1_ws;
+  uint8_t* m2p = (uint8_t*)m2_ws;
+  float* p2p = (float*)p2_ws;
+  uint8_t* idx2p = (uint8_t*)idx2_ws;
+  float* h1p = (float*)h1_ws;
+  uint8_t* m3p = (uint8_t*)m3_ws;
+  float* d3p = (float*)d3_ws;
+  float* logpp = (float*)logp_ws;
+  float* glogp = (float*)glog_ws;
+  float* gh1p = (float*)gh1_ws;
+  float* ga2p = (float*)ga2_ws;
+  float* ga1p = (float*)ga1_ws;
+  float* partp = (float*)part_ws;
+  float* lpartp = (float*)loss_part;
+  float* loutp = (float*)loss_out;
+  unsigned long long* seedp = (unsigned long long*)seed_dev;
+  float lrf = (float)lr, muf = (float)mu;
+  int do_sgd_i = do_sgd ? 1 : 0, Bi = B, tri = training ? 1 : 0;
+  void* args[] = {&xp, &tgtp, &glp, &p1p, &idx1p, &m2p, &p2p, &idx2p,
+                  &h1p, &m3p, &d3p, &logpp, &glogp, &gh1p, &ga2p, &ga1p,
+                  &partp, &lpartp, &loutp, &seedp, &prm, &grd, &buf,
+                  &lrf, &muf, &do_sgd_i, &Bi, &tri, &split, &bchunk,
+                  &nch};
+  HIP_CHECK(hipLaunchCooperativeKernel(
+      reinterpret_cast<const void*>(net_step_kernel), dim3(nblk),
+      dim3(256), args, 0, S(stream)));
+}
+
 void add_inplace(uintptr_t dst, uintptr_t src, int64_t n, int dtype,
                  uintptr_t stream) {
   if (dtype == 7) {  // ncclFloat32 numbering (dist wrapper's _DTYPE)
@@ -1733,6 +2079,8 @@ PYBIND11_MODULE(_kernels, m) {
   m.def("sgd_step", &sgd_step);
   m.def("net_fused_fwd", &net_fused_fwd);
   m.def("net_fused_bwd", &net_fused_bwd);
+  m.def("net_step", &net_step);
+  m.def("net_step_available", &net_step_available);
   m.def("add_inplace", &add_inplace);
   m.def("reduce_columns", &reduce_columns);
   m.def("scale_f32", &scale_f32);

@@ -97,6 +97,74 @@ def test_fused_training_mode_statistics():
     assert flat.abs().sum() > 0
 
 
+def test_megakernel_matches_fused_eval():
+    """Single-launch cooperative step == 6-dispatch fused path, exactly
+    (same shared __device__ code, eval mode => deterministic)."""
+    from dist_tuto_pth_amd.optim import FusedSGD
+    from dist_tuto_pth_amd.ops.fused import (net_fused_train_step,
+                                             net_step_available)
+    if not net_step_available():
+        pytest.skip("cooperative launch unavailable")
+    _, net_a, x, tgt = _mk(6)
+    _, net_b, _, _ = _mk(6)
+    xg, tg = x.to(DEV), tgt.to(DEV)
+
+    attach_flat_grads(net_a)
+    opt_a = FusedSGD(net_a.parameters(), lr=0.01, momentum=0.5)
+    loss_a = net_fused_step(net_a, xg, tg).clone()
+    opt_a.step()
+
+    attach_flat_grads(net_b)
+    opt_b = FusedSGD(net_b.parameters(), lr=0.01, momentum=0.5)
+    loss_b = net_fused_train_step(net_b, xg, tg, opt_b).clone()
+    torch.cuda.synchronize()
+
+    assert torch.allclose(loss_a, loss_b, atol=1e-6), \
+        (loss_a.item(), loss_b.item())
+    for (n, pa), pb in zip(net_a.named_parameters(), net_b.parameters()):
+        assert torch.allclose(pa, pb, atol=1e-6), \
+            (n, (pa - pb).abs().max().item())
+        assert torch.allclose(pa.grad, pb.grad, atol=1e-6), n
+    for ba, bb in zip(opt_a._bufs, opt_b._bufs):
+        assert torch.allclose(ba, bb, atol=1e-6)
+
+
+def test_megakernel_no_sgd_leaves_params():
+    """do_sgd=False (the DP mode): grads written, params untouched."""
+    from dist_tuto_pth_amd.ops.fused import (net_fused_train_step,
+                                             net_step_available)
+    if not net_step_available():
+        pytest.skip("cooperative launch unavailable")
+    _, net_g, x, tgt = _mk(7)
+    p0 = [p.clone() for p in net_g.parameters()]
+    flat = attach_flat_grads(net_g)
+    net_fused_train_step(net_g, x.to(DEV), tgt.to(DEV), do_sgd=False)
+    torch.cuda.synchronize()
+    assert flat.abs().sum() > 0
+    for a, p in zip(p0, net_g.parameters()):
+        assert torch.equal(a, p)
+
+
+def test_megakernel_training_convergence():
+    """A few single-launch steps reduce the loss on a fixed batch."""
+    from dist_tuto_pth_amd.optim import FusedSGD
+    from dist_tuto_pth_amd.ops.fused import (net_fused_train_step,
+                                             net_step_available)
+    if not net_step_available():
+        pytest.skip("cooperative launch unavailable")
+    _, net_g, x, tgt = _mk(8, B=256)
+    net_g.train()
+    attach_flat_grads(net_g)
+    opt = FusedSGD(net_g.parameters(), lr=0.05, momentum=0.5)
+    xg, tg = x.to(DEV), tgt.to(DEV)
+    losses = []
+    for _ in range(150):
+        losses.append(net_fused_train_step(net_g, xg, tg, opt).item())
+    first = sum(losses[:10]) / 10
+    last = sum(losses[-10:]) / 10
+    assert last < first - 0.12, (first, last)
+
+
 def test_fused_training_convergence():
     """A few fused steps reduce the loss on a fixed batch."""
     from dist_tuto_pth_amd.optim import FusedSGD
