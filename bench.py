@@ -90,6 +90,7 @@ def main():
     if use_fused:
         from dist_tuto_pth_amd.ops.fused import (attach_flat_grads,
                                                  net_fused_step,
+                                                 net_fused_step_opt,
                                                  net_fused_train_step,
                                                  net_step_available)
         flat_grads = attach_flat_grads(model)
@@ -107,12 +108,15 @@ def main():
                 dist.all_reduce(flat_grads, op=dist.ReduceOp.AVG)
                 opt.step()
                 return loss
+            if world == 1:
+                # SGD update fused into the combine kernel (one
+                # dispatch fewer; no all-reduce to wait for)
+                return net_fused_step_opt(model, x, tgt, opt)
             loss = net_fused_step(model, x, tgt)
-            if world > 1:
-                # one flat all-reduce with built-in averaging: the
-                # semantics of average_gradients (train_dist.py:94-100)
-                # in a single xGMI message
-                dist.all_reduce(flat_grads, op=dist.ReduceOp.AVG)
+            # one flat all-reduce with built-in averaging: the
+            # semantics of average_gradients (train_dist.py:94-100)
+            # in a single xGMI message
+            dist.all_reduce(flat_grads, op=dist.ReduceOp.AVG)
             opt.step()
             return loss
         if ddp is not None:

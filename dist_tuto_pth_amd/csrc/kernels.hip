@@ -21,6 +21,7 @@
 #include <pybind11/stl.h>
 
 #include <cstdint>
+#include <cstdlib>
 #include <stdexcept>
 #include <string>
 #include <vector>
@@ -795,8 +796,8 @@ __device__ __forceinline__ float net_fwd_sample(
     uint8_t* __restrict__ idx2_ws, float* __restrict__ h1_ws,
     uint8_t* __restrict__ m3_ws, float* __restrict__ d3_ws,
     float* __restrict__ logp_ws,
-    // LDS carve (sizes: 784, 260, 5760, 1440, 5020, 1280, 320, 50, 10)
-    float* xs, float* w1s, float* a1, float* p1, float* w2s, float* d2,
+    // LDS carve (sizes: 784, 260, 1440, 5020, 320, 50, 10)
+    float* xs, float* w1s, float* p1, float* w2s,
     float* p2, float* d3, float* logits) {
   float ret = 0.f;
   {
@@ -808,42 +809,33 @@ __device__ __forceinline__ float net_fwd_sample(
     if (tid < N_C2K) w2s[N_C2K * 250 + tid] = b2[tid];
     __syncthreads();
 
-    // conv1: 1->10, k5, 28->24.  Two outputs per loop iteration
-    // (channel halves) -> two independent 25-FMA chains in flight
-    // instead of one (the kernel is latency-bound at ~1 wave/SIMD).
-    for (int i = tid; i < N_A1 / 2; i += 256) {
-      const int j = i + N_A1 / 2;
-      const int ka = i / 576, kb = j / 576;
-      const int oha = (i / 24) % 24, owa = i % 24;
-      const int ohb = (j / 24) % 24, owb = j % 24;
-      const float* wka = w1s + ka * 25;
-      const float* wkb = w1s + kb * 25;
-      const float* xpa = xs + oha * 28 + owa;
-      const float* xpb = xs + ohb * 28 + owb;
-      float acca = w1s[N_C1K * 25 + ka];
-      float accb = w1s[N_C1K * 25 + kb];
+    // conv1 + pool1 + relu in ONE register-blocked stage: each thread
+    // owns a 2x2 pooling window, computes its four conv outputs in
+    // four independent accumulator chains (the kernel is LDS-latency
+    // bound at ~1 wave/SIMD — profiles/), then pools in registers.
+    // The 5760-float conv1 activation never touches LDS.
+    for (int i = tid; i < N_P1; i += 256) {
+      const int c = i / 144, ph = (i / 12) % 12, pw = i % 12;
+      const float* wk = w1s + c * 25;
+      const float* xp = xs + ph * 2 * 28 + pw * 2;
+      const float bias = w1s[N_C1K * 25 + c];
+      float q00 = bias, q01 = bias, q10 = bias, q11 = bias;
       #pragma unroll
       for (int r = 0; r < 5; ++r) {
         #pragma unroll
         for (int s = 0; s < 5; ++s) {
-          acca += xpa[r * 28 + s] * wka[r * 5 + s];
-          accb += xpb[r * 28 + s] * wkb[r * 5 + s];
+          const float w = wk[r * 5 + s];
+          const float* xr = xp + r * 28 + s;
+          q00 += xr[0] * w;
+          q01 += xr[1] * w;
+          q10 += xr[28] * w;
+          q11 += xr[29] * w;
         }
       }
-      a1[i] = acca;
-      a1[j] = accb;
-    }
-    __syncthreads();
-
-    // pool1 (2x2) + relu, stash p1 + idx1
-    for (int i = tid; i < N_P1; i += 256) {
-      const int c = i / 144, oh = (i / 12) % 12, ow = i % 12;
-      const float* ap = a1 + c * 576 + oh * 2 * 24 + ow * 2;
-      float v0 = ap[0], v1 = ap[1], v2 = ap[24], v3 = ap[25];
-      int am = 0; float m = v0;
-      if (v1 > m) { m = v1; am = 1; }
-      if (v2 > m) { m = v2; am = 2; }
-      if (v3 > m) { m = v3; am = 3; }
+      int am = 0; float m = q00;
+      if (q01 > m) { m = q01; am = 1; }
+      if (q10 > m) { m = q10; am = 2; }
+      if (q11 > m) { m = q11; am = 3; }
       const float o = m > 0.f ? m : 0.f;
       p1[i] = o;
       p1_ws[(int64_t)b * N_P1 + i] = o;
@@ -851,61 +843,47 @@ __device__ __forceinline__ float net_fwd_sample(
     }
     __syncthreads();
 
-    // conv2: 10->20, k5, 12->8, then channelwise dropout (conv2_drop).
-    // Two independent accumulator chains (channel parity) keep the
-    // 250-FMA reduction ILP-bound instead of LDS-latency-bound (there
-    // is ~1 wave/SIMD at reference batch sizes, so TLP cannot hide it).
-    for (int i = tid; i < N_A2; i += 256) {
-      const int k = i / 64, oh = (i / 8) % 8, ow = i % 8;
-      const float* wk = w2s + k * 250;
-      // four accumulator chains: (channel parity) x (row parity)
-      float a0 = w2s[N_C2K * 250 + k], a1 = 0.f, a2 = 0.f, a3 = 0.f;
+    // conv2 + dropout2d + pool2 + relu, same register-blocked shape:
+    // one thread per pooled cell, four conv outputs in four chains,
+    // channel dropout applied in registers before the max.
+    for (int i = tid; i < N_P2; i += 256) {
+      const int k = i / 16, ph = (i / 4) % 4, pw = i % 4;
+      const float bias = w2s[N_C2K * 250 + k];
+      float q00 = bias, q01 = bias, q10 = bias, q11 = bias;
       #pragma unroll
-      for (int c = 0; c < 10; c += 2) {
-        const float* pp0 = p1 + c * 144 + oh * 12 + ow;
-        const float* wc0 = wk + c * 25;
-        const float* pp1 = pp0 + 144;
-        const float* wc1 = wc0 + 25;
+      for (int c = 0; c < 10; ++c) {
+        const float* pp = p1 + c * 144 + ph * 2 * 12 + pw * 2;
+        const float* wc = w2s + (k * 10 + c) * 25;
         #pragma unroll
         for (int r = 0; r < 5; ++r) {
           #pragma unroll
           for (int s = 0; s < 5; ++s) {
-            if (r & 1) {
-              a2 += pp0[r * 12 + s] * wc0[r * 5 + s];
-              a3 += pp1[r * 12 + s] * wc1[r * 5 + s];
-            } else {
-              a0 += pp0[r * 12 + s] * wc0[r * 5 + s];
-              a1 += pp1[r * 12 + s] * wc1[r * 5 + s];
-            }
+            const float w = wc[r * 5 + s];
+            const float* pr = pp + r * 12 + s;
+            q00 += pr[0] * w;
+            q01 += pr[1] * w;
+            q10 += pr[12] * w;
+            q11 += pr[13] * w;
           }
         }
       }
-      float acc = (a0 + a1) + (a2 + a3);
       if (training) {
         const uint32_t rr = mix32(seed, (uint64_t)b * N_C2K + k);
-        acc = (rr >= 0x80000000u) ? acc * 2.f : 0.f;
+        const float dsc = (rr >= 0x80000000u) ? 2.f : 0.f;
+        q00 *= dsc; q01 *= dsc; q10 *= dsc; q11 *= dsc;
       }
-      d2[i] = acc;
-    }
-    if (training && tid < N_C2K) {
-      const uint32_t rr = mix32(seed, (uint64_t)b * N_C2K + tid);
-      m2_ws[(int64_t)b * N_C2K + tid] = rr >= 0x80000000u;
-    }
-    __syncthreads();
-
-    // pool2 + relu -> p2 (fc1 input), stash
-    for (int i = tid; i < N_P2; i += 256) {
-      const int c = i / 16, oh = (i / 4) % 4, ow = i % 4;
-      const float* dp = d2 + c * 64 + oh * 2 * 8 + ow * 2;
-      float v0 = dp[0], v1 = dp[1], v2 = dp[8], v3 = dp[9];
-      int am = 0; float m = v0;
-      if (v1 > m) { m = v1; am = 1; }
-      if (v2 > m) { m = v2; am = 2; }
-      if (v3 > m) { m = v3; am = 3; }
+      int am = 0; float m = q00;
+      if (q01 > m) { m = q01; am = 1; }
+      if (q10 > m) { m = q10; am = 2; }
+      if (q11 > m) { m = q11; am = 3; }
       const float o = m > 0.f ? m : 0.f;
       p2[i] = o;
       p2_ws[(int64_t)b * N_P2 + i] = o;
       idx2_ws[(int64_t)b * N_P2 + i] = (uint8_t)(m > 0.f ? am : (am | 4));
+    }
+    if (training && tid < N_C2K) {
+      const uint32_t rr = mix32(seed, (uint64_t)b * N_C2K + tid);
+      m2_ws[(int64_t)b * N_C2K + tid] = rr >= 0x80000000u;
     }
     __syncthreads();
 
@@ -999,10 +977,8 @@ net_fused_fwd_kernel(
     int B, int training) {
   __shared__ __attribute__((aligned(16))) float xs[784];
   __shared__ float w1s[N_C1K * 25 + N_C1K];
-  __shared__ float a1[N_A1];
   __shared__ float p1[N_P1];
   __shared__ float w2s[N_C2K * 10 * 25 + N_C2K];
-  __shared__ float d2[N_A2];
   __shared__ float p2[N_P2];
   __shared__ float d3[N_H1];
   __shared__ float logits[N_CLS];
@@ -1013,7 +989,7 @@ net_fused_fwd_kernel(
     const float lp_t = net_fwd_sample(
         b, tid, B, training, seed, x, w1, b1, w2, b2, wf1, bf1, wf2, bf2,
         tgt, p1_ws, idx1_ws, m2_ws, p2_ws, idx2_ws, h1_ws, m3_ws, d3_ws,
-        logp_ws, xs, w1s, a1, p1, w2s, d2, p2, d3, logits);
+        logp_ws, xs, w1s, p1, w2s, p2, d3, logits);
     if (tid == 0) atomicAdd(loss, -lp_t / B);
   }
 }
@@ -1211,13 +1187,13 @@ net_fused_bwd_kernel(
 #define OFF_BF2 21830     // 10
 #define T_CONV2 20        // ceil(5020/256)
 #define T_FC1 63          // ceil(16050/256)
-#define T_CONV1 4         // 4 sub-blocks over output-row quarters
+#define T_CONV1 8         // 8 sub-blocks over 3-output-row bands
 #define T_FC2 2           // ceil(510/256)
 #define GW_TILES (T_CONV2 + T_FC1 + T_CONV1 + T_FC2)
-// conv1's 4 sub-blocks write disjoint 260-float slices: sub 0 to the
-// canonical [OFF_W1,OFF_B1] region, subs 1-3 to an extension past
+// conv1's 8 sub-blocks write disjoint 260-float slices: sub 0 to the
+// canonical [OFF_W1,OFF_B1] region, subs 1-7 to an extension past
 // GW_TOTAL; the combine kernel folds the extension back in.
-#define GW_ROW (GW_TOTAL + 3 * 260)
+#define GW_ROW (GW_TOTAL + 7 * 260)
 
 // One (tile, batch-chunk) partial weight-gradient reduction.  Tiles
 // walk [conv2 | fc1 | conv1x4 | fc2]; partials land in my[GW_ROW]
@@ -1240,25 +1216,31 @@ __device__ __forceinline__ void net_gw_tile(
       if (i < 5000) {
         const int k = i / 250, c = (i / 25) % 10;
         const int r = (i / 5) % 5, sx = i % 5;
-        // two accumulator chains (output-row parity): the 256-FMA
-        // serial chain was latency-bound at ~1 occupancy-hidden wave
-        float ae = 0.f, ao = 0.f;
+        // four accumulator chains over output-row parity + float4
+        // vector loads of the 16B-aligned ga2 rows: the serial 256-FMA
+        // chain of scalar loads was latency-bound (VALUBusy 7%)
+        float q0 = 0.f, q1 = 0.f, q2 = 0.f, q3 = 0.f;
         for (int b = b0; b < b1; ++b) {
-          const float* gk = ga2_ws + (int64_t)b * N_A2 + k * 64;
+          const float4* g4 = reinterpret_cast<const float4*>(
+              ga2_ws + (int64_t)b * N_A2 + k * 64);
           const float* xc = p1_ws + (int64_t)b * N_P1 + c * 144 +
                             r * 12 + sx;
           #pragma unroll
-          for (int oh = 0; oh < 8; oh += 2) {
-            const float* grow = gk + oh * 8;
-            const float* xrow = xc + oh * 12;
-            #pragma unroll
-            for (int ow = 0; ow < 8; ++ow) {
-              ae += grow[ow] * xrow[ow];
-              ao += grow[8 + ow] * xrow[12 + ow];
-            }
+          for (int oh = 0; oh < 8; ++oh) {
+            const float4 ga = g4[oh * 2], gb = g4[oh * 2 + 1];
+            const float* xr = xc + oh * 12;
+            float s0, s1;
+            s0 = ga.x * xr[0] + ga.y * xr[1] + ga.z * xr[2] +
+                 ga.w * xr[3];
+            s1 = gb.x * xr[4] + gb.y * xr[5] + gb.z * xr[6] +
+                 gb.w * xr[7];
+            if ((oh & 3) == 0) q0 += s0 + s1;
+            else if ((oh & 3) == 1) q1 += s0 + s1;
+            else if ((oh & 3) == 2) q2 += s0 + s1;
+            else q3 += s0 + s1;
           }
         }
-        acc = ae + ao;
+        acc = (q0 + q1) + (q2 + q3);
         my[OFF_W2 + i] = acc;
       } else {
         const int k = i - 5000;
@@ -1306,18 +1288,18 @@ __device__ __forceinline__ void net_gw_tile(
     // 250 outputs is too little parallelism for element-per-thread at
     // this cost (24x24 window x batch): split each output over its 24
     // output rows -> 6000 independent items, LDS-atomic reduce, and
-    // over 4 sub-blocks (6 output rows each) so the grid column's
-    // straggler block shrinks 4x.  Sub 0 writes the canonical region,
-    // subs 1-3 the extension rows summed by the combine kernel.
+    // over 8 sub-blocks (3 output rows each) so the grid column's
+    // straggler block shrinks 8x.  Sub 0 writes the canonical region,
+    // subs 1-7 the extension rows summed by the combine kernel.
     const int sub = tile;
-    const int oh0 = sub * 6;
+    const int oh0 = sub * 3;
     __shared__ float wacc[260];
     for (int i = tid; i < 260; i += 256) wacc[i] = 0.f;
     __syncthreads();
-    for (int it = tid; it < 1500 + 60; it += 256) {
+    for (int it = tid; it < 750 + 30; it += 256) {
       float a = 0.f;
-      if (it < 1500) {
-        const int e = it / 6, oh = oh0 + it % 6;
+      if (it < 750) {
+        const int e = it / 3, oh = oh0 + it % 3;
         const int k = e / 25, r = (e / 5) % 5, sx = e % 5;
         for (int b = b0; b < b1; ++b) {
           const float* grow = ga1_ws + (int64_t)b * N_A1 + k * 576 +
@@ -1333,8 +1315,8 @@ __device__ __forceinline__ void net_gw_tile(
         }
         atomicAdd(&wacc[e], a);
       } else {
-        const int j = it - 1500;
-        const int k = j / 6, oh = oh0 + j % 6;
+        const int j = it - 750;
+        const int k = j / 3, oh = oh0 + j % 3;
         for (int b = b0; b < b1; ++b) {
           const float* grow = ga1_ws + (int64_t)b * N_A1 + k * 576 +
                               oh * 24;
@@ -1380,11 +1362,11 @@ net_gw_partial_kernel(const float* __restrict__ x,
                       const float* __restrict__ gh1_ws,
                       const float* __restrict__ glog_ws,
                       float* __restrict__ part,  // [nch][GW_ROW]
-                      int B, int bchunk) {
+                      int B, int bchunk, int tile_base) {
   const int b0 = blockIdx.y * bchunk;
-  net_gw_tile(blockIdx.x, threadIdx.x, b0, min(B, b0 + bchunk),
-              part + (int64_t)blockIdx.y * GW_ROW, x, p1_ws, p2_ws,
-              d3_ws, ga1_ws, ga2_ws, gh1_ws, glog_ws);
+  net_gw_tile(blockIdx.x + tile_base, threadIdx.x, b0,
+              min(B, b0 + bchunk), part + (int64_t)blockIdx.y * GW_ROW,
+              x, p1_ws, p2_ws, d3_ws, ga1_ws, ga2_ws, gh1_ws, glog_ws);
 }
 
 // combine: grads[i] = sum over chunks of part[c][i], written through
@@ -1407,9 +1389,10 @@ __device__ __forceinline__ float net_gw_combine_elem(
   if (i < 260) {  // conv1 sub-block extension rows (see GW_ROW)
     for (int c2 = 0; c2 < nch; ++c2) {
       const float* ext = part + (int64_t)c2 * GW_ROW + GW_TOTAL;
-      a0 += ext[i];
-      a1 += ext[260 + i];
-      a2 += ext[520 + i];
+      a0 += ext[i] + ext[4 * 260 + i];
+      a1 += ext[260 + i] + ext[5 * 260 + i];
+      a2 += ext[2 * 260 + i] + ext[6 * 260 + i];
+      a3 += ext[3 * 260 + i];
     }
   }
   return (a0 + a1) + (a2 + a3);
@@ -1430,6 +1413,32 @@ __global__ void net_gw_combine_kernel(const float* __restrict__ part,
     const float acc = net_gw_combine_elem(i, nch, part);
     const int t = net_gw_tensor_of(i, off);
     g.p[t][i - off[t]] = acc;
+  }
+}
+
+// combine + SGD in one dispatch (single-GPU training: there is no
+// gradient all-reduce between combine and the optimizer step, so the
+// ~4.5 us dispatch floor of the separate sgd_step_kernel is pure
+// overhead).  Writes the grad (so .grad stays inspectable), updates
+// the momentum buffer and the parameter exactly like sgd_step_kernel.
+__global__ void net_gw_combine_sgd_kernel(const float* __restrict__ part,
+                                          GwPtrs g, GwPtrs prm,
+                                          GwPtrs buf, int nch, float lr,
+                                          float mu) {
+  const int off[9] = {OFF_W1, OFF_B1, OFF_W2, OFF_B2, OFF_WF1, OFF_BF1,
+                      OFF_WF2, OFF_BF2, GW_TOTAL};
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < GW_TOTAL;
+       i += gridDim.x * blockDim.x) {
+    const float acc = net_gw_combine_elem(i, nch, part);
+    const int t = net_gw_tensor_of(i, off);
+    const int64_t j = i - off[t];
+    g.p[t][j] = acc;
+    float v = acc;
+    if (buf.p[t]) {
+      v = mu * buf.p[t][j] + acc;
+      buf.p[t][j] = v;
+    }
+    prm.p[t][j] -= lr * v;
   }
 }
 
@@ -1474,17 +1483,15 @@ net_step_kernel(
     GwPtrs prm, GwPtrs grd, GwPtrs buf,
     float lr, float mu, int do_sgd,
     int B, int training, int split, int bchunk, int nch) {
-  // LDS union: fwd carve (14,924 floats) reused by the bwd carve
-  // (11,830) and the final-phase reduction scratch; phases are
+  // LDS union: bwd carve (11,872 floats) reused by the fwd carve
+  // (7,884) and the final-phase reduction scratch; phases are
   // separated by grid barriers.
-  __shared__ __attribute__((aligned(16))) float smem[14924];
+  __shared__ __attribute__((aligned(16))) float smem[11872];
   float* xs = smem;             // 784
   float* w1s = xs + 784;        // 260
-  float* a1 = w1s + 260;        // 5760
-  float* p1 = a1 + 5760;        // 1440
+  float* p1 = w1s + 260;        // 1440
   float* w2s = p1 + 1440;       // 5020 (fwd: w2+b2)
-  float* d2 = w2s + 5020;       // 1280
-  float* p2 = d2 + 1280;        // 320
+  float* p2 = w2s + 5020;       // 320
   float* d3 = p2 + 320;         // 50
   float* logits = d3 + 50;      // 10
   float* b_w2s = smem;          // 5000 (bwd: w2 only)
@@ -1510,7 +1517,7 @@ net_step_kernel(
     const float lp_t = net_fwd_sample(
         b, tid, B, training, seed, x, w1, b1, w2, b2, wf1, bf1, wf2, bf2,
         tgt, p1_ws, idx1_ws, m2_ws, p2_ws, idx2_ws, h1_ws, m3_ws, d3_ws,
-        logp_ws, xs, w1s, a1, p1, w2s, d2, p2, d3, logits);
+        logp_ws, xs, w1s, p1, w2s, p2, d3, logits);
     if (tid == 0) lsum += -lp_t / B;
   }
   if (tid == 0) loss_part[wg] = lsum;
@@ -1882,8 +1889,14 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
                    uintptr_t gwf2, uintptr_t gbf2, int B, bool training,
                    uintptr_t stream) {
   // enough sibling workgroups per sample to fill the 256 CUs
-  int split = 1;
-  while (split < 8 && B * split < 256) split *= 2;
+  // (DTP_BWD_SPLIT=1/2/4/8 overrides, for microbenchmarks)
+  int split = 0;
+  if (const char* e = std::getenv("DTP_BWD_SPLIT")) split = std::atoi(e);
+  if (split != 1 && split != 2 && split != 4 && split != 8) {
+    // microbench (profiles/): at B=128, 512 workgroups (2/CU) beat 256
+    split = 1;
+    while (split < 8 && B * split < 512) split *= 2;
+  }
   const int nblk = grid_for((int64_t)B * split, 1);
   hipLaunchKernelGGL(net_fused_bwd_kernel, dim3(nblk), dim3(256),
                      0, S(stream), (const float*)w2, (const float*)wf1,
@@ -1914,7 +1927,8 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
                      (const float*)p1_ws, (const float*)p2_ws,
                      (const float*)d3_ws, (const float*)ga1_ws,
                      (const float*)ga2_ws, (const float*)gh1_ws,
-                     (const float*)glog_ws, (float*)part_ws, B, bchunk);
+                     (const float*)glog_ws, (float*)part_ws, B, bchunk,
+                     0);
   GwPtrs gp;
   gp.p[0] = (float*)gw1; gp.p[1] = (float*)gb1;
   gp.p[2] = (float*)gw2; gp.p[3] = (float*)gb2;
@@ -1923,6 +1937,81 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
   hipLaunchKernelGGL(net_gw_combine_kernel,
                      dim3((GW_TOTAL + 255) / 256), dim3(256), 0,
                      S(stream), (const float*)part_ws, gp, nch);
+}
+
+// net_fused_bwd + the optimizer update fused into the combine kernel
+// (single-GPU path: no all-reduce between combine and step).  Falls
+// back is the caller's job for B > 512 (throws here).
+void net_fused_bwd_sgd(uintptr_t x, uintptr_t w2, uintptr_t wf1,
+                       uintptr_t wf2, uintptr_t tgt, uintptr_t gl,
+                       uintptr_t p1_ws, uintptr_t idx1_ws,
+                       uintptr_t m2_ws, uintptr_t p2_ws,
+                       uintptr_t idx2_ws, uintptr_t h1_ws,
+                       uintptr_t m3_ws, uintptr_t d3_ws,
+                       uintptr_t logp_ws, uintptr_t glog_ws,
+                       uintptr_t gh1_ws, uintptr_t ga2_ws,
+                       uintptr_t ga1_ws, uintptr_t part_ws,
+                       const std::vector<uintptr_t>& grd_v,
+                       const std::vector<uintptr_t>& prm_v,
+                       const std::vector<uintptr_t>& buf_v,
+                       double lr, double mu, int B, bool training,
+                       uintptr_t stream) {
+  if (B > 512)
+    throw std::runtime_error("net_fused_bwd_sgd: B > 512 uses the "
+                             "adaptive gw path; call net_fused_bwd");
+  if (grd_v.size() != 8 || prm_v.size() != 8)
+    throw std::runtime_error("net_fused_bwd_sgd: expected 8 pointers");
+  int split = 0;
+  if (const char* e = std::getenv("DTP_BWD_SPLIT")) split = std::atoi(e);
+  if (split != 1 && split != 2 && split != 4 && split != 8) {
+    split = 1;
+    while (split < 8 && B * split < 512) split *= 2;
+  }
+  const int nblk = grid_for((int64_t)B * split, 1);
+  hipLaunchKernelGGL(net_fused_bwd_kernel, dim3(nblk), dim3(256),
+                     0, S(stream), (const float*)w2, (const float*)wf1,
+                     (const float*)wf2, (const int64_t*)tgt,
+                     (const float*)gl, (const uint8_t*)idx1_ws,
+                     (const uint8_t*)m2_ws, (const uint8_t*)idx2_ws,
+                     (const float*)h1_ws, (const uint8_t*)m3_ws,
+                     (const float*)logp_ws, (float*)glog_ws,
+                     (float*)gh1_ws, (float*)ga2_ws, (float*)ga1_ws, B,
+                     training ? 1 : 0, split);
+  const int bchunk = (B + 31) / 32;
+  const int nch = (B + bchunk - 1) / bchunk;
+  hipLaunchKernelGGL(net_gw_partial_kernel, dim3(GW_TILES, nch),
+                     dim3(256), 0, S(stream), (const float*)x,
+                     (const float*)p1_ws, (const float*)p2_ws,
+                     (const float*)d3_ws, (const float*)ga1_ws,
+                     (const float*)ga2_ws, (const float*)gh1_ws,
+                     (const float*)glog_ws, (float*)part_ws, B, bchunk,
+                     0);
+  GwPtrs gp{}, pp{}, bp{};
+  for (int i = 0; i < 8; ++i) {
+    gp.p[i] = (float*)grd_v[i];
+    pp.p[i] = (float*)prm_v[i];
+    bp.p[i] = (i < (int)buf_v.size()) ? (float*)buf_v[i] : nullptr;
+  }
+  hipLaunchKernelGGL(net_gw_combine_sgd_kernel,
+                     dim3((GW_TOTAL + 255) / 256), dim3(256), 0,
+                     S(stream), (const float*)part_ws, gp, pp, bp, nch,
+                     (float)lr, (float)mu);
+}
+
+// raw tile-segment launch of the partial weight-gradient kernel
+// (microbenchmarks: time [conv2 | fc1 | conv1 | fc2] separately)
+void net_gw_partial_raw(uintptr_t x, uintptr_t p1_ws, uintptr_t p2_ws,
+                        uintptr_t d3_ws, uintptr_t ga1_ws,
+                        uintptr_t ga2_ws, uintptr_t gh1_ws,
+                        uintptr_t glog_ws, uintptr_t part_ws, int B,
+                        int bchunk, int tile_base, int ntiles, int nch,
+                        uintptr_t stream) {
+  hipLaunchKernelGGL(net_gw_partial_kernel, dim3(ntiles, nch), dim3(256),
+                     0, S(stream), (const float*)x, (const float*)p1_ws,
+                     (const float*)p2_ws, (const float*)d3_ws,
+                     (const float*)ga1_ws, (const float*)ga2_ws,
+                     (const float*)gh1_ws, (const float*)glog_ws,
+                     (float*)part_ws, B, bchunk, tile_base);
 }
 
 // ---- single-launch training step (cooperative) --------------------------
@@ -2081,6 +2170,8 @@ PYBIND11_MODULE(_kernels, m) {
   m.def("net_fused_bwd", &net_fused_bwd);
   m.def("net_step", &net_step);
   m.def("net_step_available", &net_step_available);
+  m.def("net_gw_partial_raw", &net_gw_partial_raw);
+  m.def("net_fused_bwd_sgd", &net_fused_bwd_sgd);
   m.def("add_inplace", &add_inplace);
   m.def("reduce_columns", &reduce_columns);
   m.def("scale_f32", &scale_f32);

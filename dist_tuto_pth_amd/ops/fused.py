@@ -34,9 +34,9 @@ def _ws(B: int, device) -> Dict[str, torch.Tensor]:
             "p2": f(B, 320), "idx2": u8(B, 320), "h1": f(B, 50),
             "m3": u8(B, 50), "d3": f(B, 50), "logp": f(B, 10),
             "glog": f(B, 10), "gh1": f(B, 50), "ga2": f(B, 1280),
-            # partial rows are GW_ROW = 21840 + 3*260 wide (conv1's 4
+            # partial rows are GW_ROW = 21840 + 7*260 wide (conv1's 8
             # weight-grad sub-blocks write disjoint slices; kernels.hip)
-            "ga1": f(B, 5760), "part": f(32, 22620), "loss": f(()),
+            "ga1": f(B, 5760), "part": f(32, 23660), "loss": f(()),
         }
         _ws_cache[key] = w
     return w
@@ -133,6 +133,50 @@ def net_fused_step(net, x: torch.Tensor, tgt: torch.Tensor) -> torch.Tensor:
         ws["glog"].data_ptr(), ws["gh1"].data_ptr(), ws["ga2"].data_ptr(),
         ws["ga1"].data_ptr(), ws["part"].data_ptr(),
         *[p.grad.data_ptr() for p in params], B, net.training, s)
+    return ws["loss"]
+
+
+def net_fused_step_opt(net, x: torch.Tensor, tgt: torch.Tensor,
+                       opt) -> torch.Tensor:
+    """``net_fused_step`` with the SGD+momentum update fused into the
+    combine kernel (one dispatch fewer; single-GPU training only — the
+    DP path needs the all-reduce between combine and step).  ``opt``
+    must be a FusedSGD; its momentum buffers are updated in-kernel.
+    Falls back to step+opt.step() for B > 512."""
+    k = load_native("_kernels")
+    B = x.shape[0]
+    if B > 512:
+        loss = net_fused_step(net, x, tgt)
+        opt.step()
+        return loss
+    ws = _ws(B, x.device)
+    params = [net.conv1.weight, net.conv1.bias, net.conv2.weight,
+              net.conv2.bias, net.fc1.weight, net.fc1.bias,
+              net.fc2.weight, net.fc2.bias]
+    for p in params:
+        if p.grad is None:
+            p.grad = torch.empty_like(p)
+    if "one" not in ws:
+        ws["one"] = torch.ones((), device=x.device)
+    s = _stream()
+    k.net_fused_fwd(
+        x.data_ptr(), *[p.data_ptr() for p in params], tgt.data_ptr(),
+        ws["p1"].data_ptr(), ws["idx1"].data_ptr(), ws["m2"].data_ptr(),
+        ws["p2"].data_ptr(), ws["idx2"].data_ptr(), ws["h1"].data_ptr(),
+        ws["m3"].data_ptr(), ws["d3"].data_ptr(), ws["logp"].data_ptr(),
+        ws["loss"].data_ptr(), _seed_ptr(x.device), B, net.training, s)
+    bufs = [b.data_ptr() for b in opt._bufs] if opt._bufs else []
+    k.net_fused_bwd_sgd(
+        x.data_ptr(), params[2].data_ptr(), params[4].data_ptr(),
+        params[6].data_ptr(), tgt.data_ptr(), ws["one"].data_ptr(),
+        ws["p1"].data_ptr(), ws["idx1"].data_ptr(), ws["m2"].data_ptr(),
+        ws["p2"].data_ptr(), ws["idx2"].data_ptr(), ws["h1"].data_ptr(),
+        ws["m3"].data_ptr(), ws["d3"].data_ptr(), ws["logp"].data_ptr(),
+        ws["glog"].data_ptr(), ws["gh1"].data_ptr(), ws["ga2"].data_ptr(),
+        ws["ga1"].data_ptr(), ws["part"].data_ptr(),
+        [p.grad.data_ptr() for p in params],
+        [p.data_ptr() for p in params], bufs, opt.lr, opt.momentum, B,
+        net.training, s)
     return ws["loss"]
 
 

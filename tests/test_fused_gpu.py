@@ -97,6 +97,33 @@ def test_fused_training_mode_statistics():
     assert flat.abs().sum() > 0
 
 
+def test_fused_step_opt_matches_separate_sgd_eval():
+    """combine+SGD fused into one kernel == combine then sgd_step
+    (same update formula; eval mode => deterministic)."""
+    from dist_tuto_pth_amd.optim import FusedSGD
+    from dist_tuto_pth_amd.ops.fused import net_fused_step_opt
+    _, net_a, x, tgt = _mk(9)
+    _, net_b, _, _ = _mk(9)
+    xg, tg = x.to(DEV), tgt.to(DEV)
+
+    attach_flat_grads(net_a)
+    opt_a = FusedSGD(net_a.parameters(), lr=0.01, momentum=0.5)
+    loss_a = net_fused_step(net_a, xg, tg).clone()
+    opt_a.step()
+
+    attach_flat_grads(net_b)
+    opt_b = FusedSGD(net_b.parameters(), lr=0.01, momentum=0.5)
+    loss_b = net_fused_step_opt(net_b, xg, tg, opt_b).clone()
+    torch.cuda.synchronize()
+
+    assert torch.allclose(loss_a, loss_b, atol=1e-6)
+    for (n, pa), pb in zip(net_a.named_parameters(), net_b.parameters()):
+        assert torch.allclose(pa, pb, atol=1e-6), n
+        assert torch.allclose(pa.grad, pb.grad, atol=1e-6), n
+    for ba, bb in zip(opt_a._bufs, opt_b._bufs):
+        assert torch.allclose(ba, bb, atol=1e-6)
+
+
 def test_megakernel_matches_fused_eval():
     """Single-launch cooperative step == 6-dispatch fused path, exactly
     (same shared __device__ code, eval mode => deterministic)."""

@@ -145,3 +145,13 @@ def test_net_param_count_matches_reference():
     from dist_tuto_pth_amd.models import Net
     n = sum(p.numel() for p in Net().parameters())
     assert n == 21840
+
+
+def test_net_step_available_false_on_cpu():
+    """The single-launch cooperative step kernel reports unavailable on
+    a CPU-only machine instead of raising (the GPU paths fail loudly,
+    the capability probe does not)."""
+    from dist_tuto_pth_amd.ops.fused import net_step_available
+    import torch
+    if not torch.cuda.is_available():
+        assert net_step_available() is False
