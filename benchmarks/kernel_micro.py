@@ -22,7 +22,7 @@ from dist_tuto_pth_amd.ops.fused import _ws, attach_flat_grads  # noqa: E402
 from dist_tuto_pth_amd.utils.native import load_native  # noqa: E402
 
 # tile segment table — keep in sync with csrc/kernels.hip
-T_CONV2, T_FC1, T_CONV1, T_FC2 = 20, 63, 4, 2
+T_CONV2, T_FC1, T_CONV1, T_FC2 = 20, 63, 8, 2
 
 
 def time_fn(fn, reps=200, warmup=20):
