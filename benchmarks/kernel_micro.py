@@ -71,7 +71,7 @@ def main():
                         ws["idx2"].data_ptr(), ws["h1"].data_ptr(),
                         ws["m3"].data_ptr(), ws["d3"].data_ptr(),
                         ws["logp"].data_ptr(), ws["loss"].data_ptr(),
-                        _seed_ptr(dev), B, True, s)
+                        0, _seed_ptr(dev), B, True, s)
 
     def bwd():
         k.net_fused_bwd(x.data_ptr(), pp[2], pp[4], pp[6], tgt.data_ptr(),
@@ -83,7 +83,8 @@ def main():
                         ws["glog"].data_ptr(), ws["gh1"].data_ptr(),
                         ws["ga2"].data_ptr(), ws["ga1"].data_ptr(),
                         ws["part"].data_ptr(),
-                        *[p.grad.data_ptr() for p in params], B, True, s)
+                        *[p.grad.data_ptr() for p in params], B, True,
+                        0, 0, 0, s)
 
     fwd()
     bwd()

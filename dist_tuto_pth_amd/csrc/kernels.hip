@@ -972,7 +972,11 @@ net_fused_fwd_kernel(
     uint8_t* __restrict__ m3_ws,      // [B,50]
     float* __restrict__ d3_ws,        // [B,50]
     float* __restrict__ logp_ws,      // [B,10]
-    float* __restrict__ loss,         // scalar (pre-zeroed)
+    float* __restrict__ loss,         // scalar (pre-zeroed) — legacy mode
+    float* __restrict__ loss_part,    // [gridDim.x] or null: when set,
+                                      // per-block partials (no atomics,
+                                      // no pre-zeroing; the combine
+                                      // kernel finalizes the loss)
     const unsigned long long* __restrict__ seed_p,
     int B, int training) {
   __shared__ __attribute__((aligned(16))) float xs[784];
@@ -985,12 +989,17 @@ net_fused_fwd_kernel(
   const int tid = threadIdx.x;
   const uint64_t seed = seed_p[0];
 
+  float lsum = 0.f;
   for (int b = blockIdx.x; b < B; b += gridDim.x) {
     const float lp_t = net_fwd_sample(
         b, tid, B, training, seed, x, w1, b1, w2, b2, wf1, bf1, wf2, bf2,
         tgt, p1_ws, idx1_ws, m2_ws, p2_ws, idx2_ws, h1_ws, m3_ws, d3_ws,
         logp_ws, xs, w1s, p1, w2s, p2, d3, logits);
-    if (tid == 0) atomicAdd(loss, -lp_t / B);
+    if (tid == 0) lsum += -lp_t / B;
+  }
+  if (tid == 0) {
+    if (loss_part) loss_part[blockIdx.x] = lsum;
+    else atomicAdd(loss, lsum);
   }
 }
 
@@ -1084,43 +1093,54 @@ __device__ __forceinline__ void net_bwd_sample(
     }
     __syncthreads();
 
-    // conv2 bwd_x through the padded tile (fixed 5x5 bounds, two
-    // independent accumulator chains) + pool1 bwd: each pooled
-    // position owns its 2x2 window exclusively, so all four slots are
-    // written here — no zeroing pass.  Partitioned across the `split`
-    // sibling workgroups (N_P1=1440 divisible by 1/2/4/8).
-    const int seg = N_P1 / split;
-    const int i_end = (half + 1) * seg;
-    for (int i = half * seg + tid; i < i_end; i += 256) {
-      const int c = i / 144, h = (i / 12) % 12, wc = i % 12;
-      float a0 = 0.f, a1 = 0.f;
-      for (int k = 0; k < N_C2K; k += 2) {
-        const float* g0 = gd2p + k * 256;
-        const float* g1 = g0 + 256;
-        const float* w0 = w2s + (k * 10 + c) * 25;
-        const float* w1v = w0 + 250;
+    // conv2 bwd_x through the padded tile + pool1 bwd, register-blocked
+    // 4-wide: each thread owns FOUR horizontally-adjacent pooled
+    // positions (one accumulator chain each); a (k, r) row contributes
+    // 8 consecutive gd2p values shared by all four outputs, cutting
+    // LDS reads ~2.5x vs one-output-per-thread (the kernel is
+    // LDS-latency-bound at this occupancy — profiles/).  Each pooled
+    // position owns its 2x2 ga1 window exclusively, so all four slots
+    // are written with no zeroing pass.  Partitioned across the
+    // `split` sibling workgroups (360 strips divisible by 1/2/4/8).
+    const int seg = (N_P1 / 4) / split;
+    const int t_end = (half + 1) * seg;
+    for (int t = half * seg + tid; t < t_end; t += 256) {
+      const int c = t / 36, rem = t % 36;
+      const int h = rem / 3, wc0 = (rem % 3) * 4;
+      float q0 = 0.f, q1 = 0.f, q2 = 0.f, q3 = 0.f;
+      for (int k = 0; k < N_C2K; ++k) {
+        const float* gk = gd2p + k * 256;
+        const float* wk = w2s + (k * 10 + c) * 25;
         #pragma unroll
         for (int r = 0; r < 5; ++r) {
-          const float* gr0 = g0 + (h - r + 4) * 16 + (wc + 4);
-          const float* gr1 = g1 + (h - r + 4) * 16 + (wc + 4);
-          #pragma unroll
-          for (int s = 0; s < 5; ++s) {
-            a0 += gr0[-s] * w0[r * 5 + s];
-            a1 += gr1[-s] * w1v[r * 5 + s];
-          }
+          const float* row = gk + (h - r + 4) * 16 + wc0;
+          const float w0 = wk[r * 5 + 0], w1 = wk[r * 5 + 1];
+          const float w2v = wk[r * 5 + 2], w3 = wk[r * 5 + 3];
+          const float w4 = wk[r * 5 + 4];
+          const float e0 = row[0], e1 = row[1], e2 = row[2];
+          const float e3 = row[3], e4 = row[4], e5 = row[5];
+          const float e6 = row[6], e7 = row[7];
+          q0 += e4 * w0 + e3 * w1 + e2 * w2v + e1 * w3 + e0 * w4;
+          q1 += e5 * w0 + e4 * w1 + e3 * w2v + e2 * w3 + e1 * w4;
+          q2 += e6 * w0 + e5 * w1 + e4 * w2v + e3 * w3 + e2 * w4;
+          q3 += e7 * w0 + e6 * w1 + e5 * w2v + e4 * w3 + e3 * w4;
         }
       }
-      const float acc = a0 + a1;
-      const uint8_t v = idx1_ws[(int64_t)b * N_P1 + i];
-      const int am = v & 3;
-      const float g = (v & 4) ? 0.f : acc;
-      const int oh = (i / 12) % 12, ow = i % 12;
-      float* gp = ga1_ws + (int64_t)b * N_A1 + c * 576 +
-                  oh * 2 * 24 + ow * 2;
-      gp[0] = am == 0 ? g : 0.f;
-      gp[1] = am == 1 ? g : 0.f;
-      gp[24] = am == 2 ? g : 0.f;
-      gp[25] = am == 3 ? g : 0.f;
+      const float qq[4] = {q0, q1, q2, q3};
+      const int64_t ib = (int64_t)b * N_P1 + c * 144 + h * 12 + wc0;
+      float* gp_base = ga1_ws + (int64_t)b * N_A1 + c * 576 +
+                       h * 2 * 24 + wc0 * 2;
+      #pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const uint8_t v = idx1_ws[ib + j];
+        const int am = v & 3;
+        const float g = (v & 4) ? 0.f : qq[j];
+        float* gp = gp_base + j * 2;
+        gp[0] = am == 0 ? g : 0.f;
+        gp[1] = am == 1 ? g : 0.f;
+        gp[24] = am == 2 ? g : 0.f;
+        gp[25] = am == 3 ? g : 0.f;
+      }
     }
     __syncthreads();
   }
@@ -1210,48 +1230,48 @@ __device__ __forceinline__ void net_gw_tile(
     const float* __restrict__ gh1_ws,
     const float* __restrict__ glog_ws) {
   if (tile < T_CONV2) {  // conv2: gw [20][10][5][5] + gb [20]
-    const int i = tile * 256 + tid;
-    if (i < 5020) {
-      float acc = 0.f;
-      if (i < 5000) {
-        const int k = i / 250, c = (i / 25) % 10;
-        const int r = (i / 5) % 5, sx = i % 5;
-        // four accumulator chains over output-row parity + float4
-        // vector loads of the 16B-aligned ga2 rows: the serial 256-FMA
-        // chain of scalar loads was latency-bound (VALUBusy 7%)
-        float q0 = 0.f, q1 = 0.f, q2 = 0.f, q3 = 0.f;
-        for (int b = b0; b < b1; ++b) {
-          const float4* g4 = reinterpret_cast<const float4*>(
-              ga2_ws + (int64_t)b * N_A2 + k * 64);
-          const float* xc = p1_ws + (int64_t)b * N_P1 + c * 144 +
-                            r * 12 + sx;
-          #pragma unroll
-          for (int oh = 0; oh < 8; ++oh) {
-            const float4 ga = g4[oh * 2], gb = g4[oh * 2 + 1];
-            const float* xr = xc + oh * 12;
-            float s0, s1;
-            s0 = ga.x * xr[0] + ga.y * xr[1] + ga.z * xr[2] +
-                 ga.w * xr[3];
-            s1 = gb.x * xr[4] + gb.y * xr[5] + gb.z * xr[6] +
-                 gb.w * xr[7];
-            if ((oh & 3) == 0) q0 += s0 + s1;
-            else if ((oh & 3) == 1) q1 += s0 + s1;
-            else if ((oh & 3) == 2) q2 += s0 + s1;
-            else q3 += s0 + s1;
-          }
+    // one block per output channel k: stage the batch element's full
+    // p1 plane (1440 floats) and its ga2 row (64) through LDS once,
+    // then the 250 weight threads reduce from LDS (each p1 value was
+    // being re-read ~11x from L1 in the element-per-thread form).
+    // Four accumulator chains over output-row parity.
+    const int k = tile;
+    __shared__ float sp1[N_P1];
+    __shared__ float sg2[64];
+    const int c = tid / 25, r = (tid / 5) % 5, sx = tid % 5;
+    float q0 = 0.f, q1 = 0.f, q2 = 0.f, q3 = 0.f;
+    float be = 0.f, bo = 0.f;
+    for (int b = b0; b < b1; ++b) {
+      __syncthreads();
+      for (int i = tid; i < N_P1; i += 256)
+        sp1[i] = p1_ws[(int64_t)b * N_P1 + i];
+      if (tid < 64) sg2[tid] = ga2_ws[(int64_t)b * N_A2 + k * 64 + tid];
+      __syncthreads();
+      if (tid < 250) {
+        const float* xc = sp1 + c * 144 + r * 12 + sx;
+        #pragma unroll
+        for (int oh = 0; oh < 8; ++oh) {
+          const float* gr = sg2 + oh * 8;
+          const float* xr = xc + oh * 12;
+          const float s0 = gr[0] * xr[0] + gr[1] * xr[1] +
+                           gr[2] * xr[2] + gr[3] * xr[3];
+          const float s1 = gr[4] * xr[4] + gr[5] * xr[5] +
+                           gr[6] * xr[6] + gr[7] * xr[7];
+          if ((oh & 3) == 0) q0 += s0 + s1;
+          else if ((oh & 3) == 1) q1 += s0 + s1;
+          else if ((oh & 3) == 2) q2 += s0 + s1;
+          else q3 += s0 + s1;
         }
-        acc = (q0 + q1) + (q2 + q3);
-        my[OFF_W2 + i] = acc;
-      } else {
-        const int k = i - 5000;
-        for (int b = b0; b < b1; ++b) {
-          const float* gk = ga2_ws + (int64_t)b * N_A2 + k * 64;
-          #pragma unroll
-          for (int j = 0; j < 64; ++j) acc += gk[j];
+      } else if (tid == 250) {  // bias: two chains over the 64 values
+        #pragma unroll
+        for (int j = 0; j < 64; j += 2) {
+          be += sg2[j];
+          bo += sg2[j + 1];
         }
-        my[OFF_B2 + k] = acc;
       }
     }
+    if (tid < 250) my[OFF_W2 + k * 250 + tid] = (q0 + q1) + (q2 + q3);
+    if (tid == 250) my[OFF_B2 + k] = be + bo;
     return;
   }
   tile -= T_CONV2;
@@ -1404,8 +1424,35 @@ __device__ __forceinline__ int net_gw_tensor_of(int i, const int* off) {
   return t;
 }
 
+// finalize the per-block loss partials written by the forward kernel
+// (loss_part mode) and advance the dropout seed for the NEXT step —
+// runs in block 0 of a combine kernel, replacing the step-prologue
+// dispatch (~4.5 us device floor) entirely.
+__device__ __forceinline__ void net_loss_finalize(
+    const float* __restrict__ loss_part, float* __restrict__ loss_out,
+    int nblk_fwd, unsigned long long* seed_bump) {
+  __shared__ float red[256];
+  float v = 0.f;
+  for (int i = threadIdx.x; i < nblk_fwd; i += 256) v += loss_part[i];
+  red[threadIdx.x] = v;
+  __syncthreads();
+  #pragma unroll
+  for (int s2 = 128; s2 > 0; s2 >>= 1) {
+    if (threadIdx.x < s2) red[threadIdx.x] += red[threadIdx.x + s2];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    *loss_out = red[0];
+    if (seed_bump) *seed_bump += 0x9E3779B97F4A7C15ull;
+  }
+}
+
 __global__ void net_gw_combine_kernel(const float* __restrict__ part,
-                                      GwPtrs g, int nch) {
+                                      GwPtrs g, int nch,
+                                      const float* __restrict__ loss_part,
+                                      float* __restrict__ loss_out,
+                                      int nblk_fwd,
+                                      unsigned long long* seed_bump) {
   const int off[9] = {OFF_W1, OFF_B1, OFF_W2, OFF_B2, OFF_WF1, OFF_BF1,
                       OFF_WF2, OFF_BF2, GW_TOTAL};
   for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < GW_TOTAL;
@@ -1414,6 +1461,8 @@ __global__ void net_gw_combine_kernel(const float* __restrict__ part,
     const int t = net_gw_tensor_of(i, off);
     g.p[t][i - off[t]] = acc;
   }
+  if (loss_part && blockIdx.x == 0)
+    net_loss_finalize(loss_part, loss_out, nblk_fwd, seed_bump);
 }
 
 // combine + SGD in one dispatch (single-GPU training: there is no
@@ -1424,7 +1473,11 @@ __global__ void net_gw_combine_kernel(const float* __restrict__ part,
 __global__ void net_gw_combine_sgd_kernel(const float* __restrict__ part,
                                           GwPtrs g, GwPtrs prm,
                                           GwPtrs buf, int nch, float lr,
-                                          float mu) {
+                                          float mu,
+                                          const float* __restrict__ loss_part,
+                                          float* __restrict__ loss_out,
+                                          int nblk_fwd,
+                                          unsigned long long* seed_bump) {
   const int off[9] = {OFF_W1, OFF_B1, OFF_W2, OFF_B2, OFF_WF1, OFF_BF1,
                       OFF_WF2, OFF_BF2, GW_TOTAL};
   for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < GW_TOTAL;
@@ -1440,6 +1493,8 @@ __global__ void net_gw_combine_sgd_kernel(const float* __restrict__ part,
     }
     prm.p[t][j] -= lr * v;
   }
+  if (loss_part && blockIdx.x == 0)
+    net_loss_finalize(loss_part, loss_out, nblk_fwd, seed_bump);
 }
 
 // ===========================================================================
@@ -1818,11 +1873,18 @@ void net_fused_fwd(uintptr_t x, uintptr_t w1, uintptr_t b1, uintptr_t w2,
                    uintptr_t p1_ws, uintptr_t idx1_ws, uintptr_t m2_ws,
                    uintptr_t p2_ws, uintptr_t idx2_ws, uintptr_t h1_ws,
                    uintptr_t m3_ws, uintptr_t d3_ws, uintptr_t logp_ws,
-                   uintptr_t loss, uintptr_t seed_dev, int B,
+                   uintptr_t loss, uintptr_t loss_part,
+                   uintptr_t seed_dev, int B,
                    bool training, uintptr_t stream) {
-  hipLaunchKernelGGL(step_prologue_kernel, dim3(1), dim3(64), 0, S(stream),
-                     training ? (unsigned long long*)seed_dev : nullptr,
-                     (float*)loss);
+  // legacy mode (loss_part == 0): prologue dispatch zeroes the loss
+  // scalar and bumps the seed, fwd accumulates the loss atomically.
+  // loss_part mode: no prologue — per-block partials, finalized (and
+  // seed bumped) by the combine kernel of the same step.
+  if (!loss_part)
+    hipLaunchKernelGGL(step_prologue_kernel, dim3(1), dim3(64), 0,
+                       S(stream),
+                       training ? (unsigned long long*)seed_dev : nullptr,
+                       (float*)loss);
   hipLaunchKernelGGL(net_fused_fwd_kernel, dim3(grid_for(B, 1)), dim3(256),
                      0, S(stream), (const float*)x, (const float*)w1,
                      (const float*)b1, (const float*)w2, (const float*)b2,
@@ -1831,7 +1893,7 @@ void net_fused_fwd(uintptr_t x, uintptr_t w1, uintptr_t b1, uintptr_t w2,
                      (const int64_t*)tgt, (float*)p1_ws, (uint8_t*)idx1_ws,
                      (uint8_t*)m2_ws, (float*)p2_ws, (uint8_t*)idx2_ws,
                      (float*)h1_ws, (uint8_t*)m3_ws, (float*)d3_ws,
-                     (float*)logp_ws, (float*)loss,
+                     (float*)logp_ws, (float*)loss, (float*)loss_part,
                      (const unsigned long long*)seed_dev, B,
                      training ? 1 : 0);
 }
@@ -1887,15 +1949,17 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
                    uintptr_t gw1, uintptr_t gb1, uintptr_t gw2,
                    uintptr_t gb2, uintptr_t gwf1, uintptr_t gbf1,
                    uintptr_t gwf2, uintptr_t gbf2, int B, bool training,
-                   uintptr_t stream) {
+                   uintptr_t loss_part, uintptr_t loss_out,
+                   uintptr_t seed_dev, uintptr_t stream) {
   // enough sibling workgroups per sample to fill the 256 CUs
   // (DTP_BWD_SPLIT=1/2/4/8 overrides, for microbenchmarks)
   int split = 0;
   if (const char* e = std::getenv("DTP_BWD_SPLIT")) split = std::atoi(e);
   if (split != 1 && split != 2 && split != 4 && split != 8) {
-    // microbench (profiles/): at B=128, 512 workgroups (2/CU) beat 256
+    // microbench (profiles/): with the 4-wide-blocked hot loop,
+    // split=2 at B=128 (256 workgroups, 180 strips each) is fastest
     split = 1;
-    while (split < 8 && B * split < 512) split *= 2;
+    while (split < 8 && B * split < 256) split *= 2;
   }
   const int nblk = grid_for((int64_t)B * split, 1);
   hipLaunchKernelGGL(net_fused_bwd_kernel, dim3(nblk), dim3(256),
@@ -1910,7 +1974,11 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
   if (B > 512) {
     // large batch: per-op adaptive chunked reductions amortize their
     // memsets; the 32-chunk partial scheme would serialize too much
-    // work per thread
+    // work per thread.  loss_part mode is a B<=512 contract (the
+    // combine kernel is what finalizes it).
+    if (loss_part)
+      throw std::runtime_error("net_fused_bwd: loss_part mode needs "
+                               "B <= 512");
     launch_conv_gw(x, ga1_ws, gw1, gb1, B, 1, 28, 28, N_C1K, stream);
     launch_conv_gw(p1_ws, ga2_ws, gw2, gb2, B, 10, 12, 12, N_C2K,
                    stream);
@@ -1936,7 +2004,11 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
   gp.p[6] = (float*)gwf2; gp.p[7] = (float*)gbf2;
   hipLaunchKernelGGL(net_gw_combine_kernel,
                      dim3((GW_TOTAL + 255) / 256), dim3(256), 0,
-                     S(stream), (const float*)part_ws, gp, nch);
+                     S(stream), (const float*)part_ws, gp, nch,
+                     (const float*)loss_part, (float*)loss_out,
+                     grid_for(B, 1),
+                     (loss_part && training)
+                         ? (unsigned long long*)seed_dev : nullptr);
 }
 
 // net_fused_bwd + the optimizer update fused into the combine kernel
@@ -1955,7 +2027,8 @@ void net_fused_bwd_sgd(uintptr_t x, uintptr_t w2, uintptr_t wf1,
                        const std::vector<uintptr_t>& prm_v,
                        const std::vector<uintptr_t>& buf_v,
                        double lr, double mu, int B, bool training,
-                       uintptr_t stream) {
+                       uintptr_t loss_part, uintptr_t loss_out,
+                       uintptr_t seed_dev, uintptr_t stream) {
   if (B > 512)
     throw std::runtime_error("net_fused_bwd_sgd: B > 512 uses the "
                              "adaptive gw path; call net_fused_bwd");
@@ -1965,7 +2038,7 @@ void net_fused_bwd_sgd(uintptr_t x, uintptr_t w2, uintptr_t wf1,
   if (const char* e = std::getenv("DTP_BWD_SPLIT")) split = std::atoi(e);
   if (split != 1 && split != 2 && split != 4 && split != 8) {
     split = 1;
-    while (split < 8 && B * split < 512) split *= 2;
+    while (split < 8 && B * split < 256) split *= 2;
   }
   const int nblk = grid_for((int64_t)B * split, 1);
   hipLaunchKernelGGL(net_fused_bwd_kernel, dim3(nblk), dim3(256),
@@ -1995,7 +2068,10 @@ void net_fused_bwd_sgd(uintptr_t x, uintptr_t w2, uintptr_t wf1,
   hipLaunchKernelGGL(net_gw_combine_sgd_kernel,
                      dim3((GW_TOTAL + 255) / 256), dim3(256), 0,
                      S(stream), (const float*)part_ws, gp, pp, bp, nch,
-                     (float)lr, (float)mu);
+                     (float)lr, (float)mu, (const float*)loss_part,
+                     (float*)loss_out, grid_for(B, 1),
+                     (loss_part && training)
+                         ? (unsigned long long*)seed_dev : nullptr);
 }
 
 // raw tile-segment launch of the partial weight-gradient kernel

@@ -71,7 +71,7 @@ class _NetFusedLoss(torch.autograd.Function):
             ws["p1"].data_ptr(), ws["idx1"].data_ptr(), ws["m2"].data_ptr(),
             ws["p2"].data_ptr(), ws["idx2"].data_ptr(), ws["h1"].data_ptr(),
             ws["m3"].data_ptr(), ws["d3"].data_ptr(), ws["logp"].data_ptr(),
-            ws["loss"].data_ptr(), _seed_ptr(x.device), B, training,
+            ws["loss"].data_ptr(), 0, _seed_ptr(x.device), B, training,
             _stream())
         ctx.save_for_backward(x, w2, wf1, wf2, tgt)
         ctx.meta = (B, training)
@@ -97,7 +97,8 @@ class _NetFusedLoss(torch.autograd.Function):
             ws["glog"].data_ptr(), ws["gh1"].data_ptr(),
             ws["ga2"].data_ptr(), ws["ga1"].data_ptr(),
             ws["part"].data_ptr(),
-            *[g.data_ptr() for g in grads], B, training, _stream())
+            *[g.data_ptr() for g in grads], B, training, 0, 0, 0,
+            _stream())
         return (None, *grads, None, None)
 
 
@@ -117,13 +118,20 @@ def net_fused_step(net, x: torch.Tensor, tgt: torch.Tensor) -> torch.Tensor:
             p.grad = torch.empty_like(p)
     if "one" not in ws:
         ws["one"] = torch.ones((), device=x.device)
+    if "loss_part" not in ws:
+        ws["loss_part"] = torch.empty(512, device=x.device)
     s = _stream()
+    # loss_part mode (B<=512): no prologue dispatch — the forward
+    # writes per-block loss partials; the combine kernel finalizes the
+    # loss scalar and bumps the dropout seed for the next step
+    lp = ws["loss_part"].data_ptr() if B <= 512 else 0
     k.net_fused_fwd(
         x.data_ptr(), *[p.data_ptr() for p in params], tgt.data_ptr(),
         ws["p1"].data_ptr(), ws["idx1"].data_ptr(), ws["m2"].data_ptr(),
         ws["p2"].data_ptr(), ws["idx2"].data_ptr(), ws["h1"].data_ptr(),
         ws["m3"].data_ptr(), ws["d3"].data_ptr(), ws["logp"].data_ptr(),
-        ws["loss"].data_ptr(), _seed_ptr(x.device), B, net.training, s)
+        ws["loss"].data_ptr(), lp, _seed_ptr(x.device), B,
+        net.training, s)
     k.net_fused_bwd(
         x.data_ptr(), params[2].data_ptr(), params[4].data_ptr(),
         params[6].data_ptr(), tgt.data_ptr(), ws["one"].data_ptr(),
@@ -132,7 +140,8 @@ def net_fused_step(net, x: torch.Tensor, tgt: torch.Tensor) -> torch.Tensor:
         ws["m3"].data_ptr(), ws["d3"].data_ptr(), ws["logp"].data_ptr(),
         ws["glog"].data_ptr(), ws["gh1"].data_ptr(), ws["ga2"].data_ptr(),
         ws["ga1"].data_ptr(), ws["part"].data_ptr(),
-        *[p.grad.data_ptr() for p in params], B, net.training, s)
+        *[p.grad.data_ptr() for p in params], B, net.training,
+        lp, ws["loss"].data_ptr(), _seed_ptr(x.device), s)
     return ws["loss"]
 
 
@@ -158,13 +167,17 @@ def net_fused_step_opt(net, x: torch.Tensor, tgt: torch.Tensor,
             p.grad = torch.empty_like(p)
     if "one" not in ws:
         ws["one"] = torch.ones((), device=x.device)
+    if "loss_part" not in ws:
+        ws["loss_part"] = torch.empty(512, device=x.device)
     s = _stream()
+    lp = ws["loss_part"].data_ptr()
     k.net_fused_fwd(
         x.data_ptr(), *[p.data_ptr() for p in params], tgt.data_ptr(),
         ws["p1"].data_ptr(), ws["idx1"].data_ptr(), ws["m2"].data_ptr(),
         ws["p2"].data_ptr(), ws["idx2"].data_ptr(), ws["h1"].data_ptr(),
         ws["m3"].data_ptr(), ws["d3"].data_ptr(), ws["logp"].data_ptr(),
-        ws["loss"].data_ptr(), _seed_ptr(x.device), B, net.training, s)
+        ws["loss"].data_ptr(), lp, _seed_ptr(x.device), B,
+        net.training, s)
     bufs = [b.data_ptr() for b in opt._bufs] if opt._bufs else []
     k.net_fused_bwd_sgd(
         x.data_ptr(), params[2].data_ptr(), params[4].data_ptr(),
@@ -176,7 +189,7 @@ def net_fused_step_opt(net, x: torch.Tensor, tgt: torch.Tensor,
         ws["ga1"].data_ptr(), ws["part"].data_ptr(),
         [p.grad.data_ptr() for p in params],
         [p.data_ptr() for p in params], bufs, opt.lr, opt.momentum, B,
-        net.training, s)
+        net.training, lp, ws["loss"].data_ptr(), _seed_ptr(x.device), s)
     return ws["loss"]
 
 
