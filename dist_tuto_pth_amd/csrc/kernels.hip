@@ -1056,12 +1056,21 @@ __device__ __forceinline__ void net_bwd_sample(
     }
     __syncthreads();
 
-    // g_p2 = wf1^T g_h1pre  (320 outputs x 50)
+    // g_p2 = wf1^T g_h1pre  (320 outputs x 50), four accumulator
+    // chains over the reduction dim (wf1 rows are coalesced global
+    // reads; the single 50-deep load+FMA chain was latency-bound)
     for (int i = tid; i < N_P2; i += 256) {
-      float acc = 0.f;
-      for (int n = 0; n < N_H1; ++n)
-        acc += wf1[n * N_P2 + i] * gh1[n];
-      gp2[i] = acc;
+      float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+      #pragma unroll
+      for (int n = 0; n < 48; n += 4) {
+        a0 += wf1[n * N_P2 + i] * gh1[n];
+        a1 += wf1[(n + 1) * N_P2 + i] * gh1[n + 1];
+        a2 += wf1[(n + 2) * N_P2 + i] * gh1[n + 2];
+        a3 += wf1[(n + 3) * N_P2 + i] * gh1[n + 3];
+      }
+      a0 += wf1[48 * N_P2 + i] * gh1[48];
+      a1 += wf1[49 * N_P2 + i] * gh1[49];
+      gp2[i] = (a0 + a1) + (a2 + a3);
     }
     __syncthreads();
 
@@ -1418,6 +1427,32 @@ __device__ __forceinline__ float net_gw_combine_elem(
   return (a0 + a1) + (a2 + a3);
 }
 
+// quad-lane variant: four consecutive lanes cooperate on one flat-grad
+// element (lane q strides the chunk rows), combined with two xor
+// shuffles.  Quadruples the wave count of the combine kernels — at 86
+// one-element-per-thread blocks they left 2/3 of the SIMDs empty and
+// ran a latency-exposed 8-deep L2 chain (profiles/).  Returns the full
+// sum on EVERY lane of the quad (butterfly reduction).
+__device__ __forceinline__ float net_gw_combine_elem_quad(
+    int i, int q, int nch, const float* __restrict__ part) {
+  float a0 = 0.f, a1 = 0.f;
+  for (int c = q; c < nch; c += 8)
+    a0 += part[(int64_t)c * GW_ROW + i];
+  for (int c = q + 4; c < nch; c += 8)
+    a1 += part[(int64_t)c * GW_ROW + i];
+  if (i < 260 && q == 0) {  // conv1 extension rows, lane 0 of the quad
+    for (int c2 = 0; c2 < nch; ++c2) {
+      const float* ext = part + (int64_t)c2 * GW_ROW + GW_TOTAL;
+      a0 += ext[i] + ext[260 + i] + ext[2 * 260 + i] + ext[3 * 260 + i];
+      a1 += ext[4 * 260 + i] + ext[5 * 260 + i] + ext[6 * 260 + i];
+    }
+  }
+  float acc = a0 + a1;
+  acc += __shfl_xor(acc, 1, 4);
+  acc += __shfl_xor(acc, 2, 4);
+  return acc;
+}
+
 __device__ __forceinline__ int net_gw_tensor_of(int i, const int* off) {
   int t = 0;
   while (i >= off[t + 1]) ++t;
@@ -1455,11 +1490,15 @@ __global__ void net_gw_combine_kernel(const float* __restrict__ part,
                                       unsigned long long* seed_bump) {
   const int off[9] = {OFF_W1, OFF_B1, OFF_W2, OFF_B2, OFF_WF1, OFF_BF1,
                       OFF_WF2, OFF_BF2, GW_TOTAL};
-  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < GW_TOTAL;
-       i += gridDim.x * blockDim.x) {
-    const float acc = net_gw_combine_elem(i, nch, part);
-    const int t = net_gw_tensor_of(i, off);
-    g.p[t][i - off[t]] = acc;
+  for (int64_t t4 = blockIdx.x * blockDim.x + threadIdx.x;
+       t4 < (int64_t)GW_TOTAL * 4;
+       t4 += (int64_t)gridDim.x * blockDim.x) {
+    const int i = (int)(t4 >> 2), q = (int)(t4 & 3);
+    const float acc = net_gw_combine_elem_quad(i, q, nch, part);
+    if (q == 0) {
+      const int t = net_gw_tensor_of(i, off);
+      g.p[t][i - off[t]] = acc;
+    }
   }
   if (loss_part && blockIdx.x == 0)
     net_loss_finalize(loss_part, loss_out, nblk_fwd, seed_bump);
@@ -1480,18 +1519,22 @@ __global__ void net_gw_combine_sgd_kernel(const float* __restrict__ part,
                                           unsigned long long* seed_bump) {
   const int off[9] = {OFF_W1, OFF_B1, OFF_W2, OFF_B2, OFF_WF1, OFF_BF1,
                       OFF_WF2, OFF_BF2, GW_TOTAL};
-  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < GW_TOTAL;
-       i += gridDim.x * blockDim.x) {
-    const float acc = net_gw_combine_elem(i, nch, part);
-    const int t = net_gw_tensor_of(i, off);
-    const int64_t j = i - off[t];
-    g.p[t][j] = acc;
-    float v = acc;
-    if (buf.p[t]) {
-      v = mu * buf.p[t][j] + acc;
-      buf.p[t][j] = v;
+  for (int64_t t4 = blockIdx.x * blockDim.x + threadIdx.x;
+       t4 < (int64_t)GW_TOTAL * 4;
+       t4 += (int64_t)gridDim.x * blockDim.x) {
+    const int i = (int)(t4 >> 2), q = (int)(t4 & 3);
+    const float acc = net_gw_combine_elem_quad(i, q, nch, part);
+    if (q == 0) {
+      const int t = net_gw_tensor_of(i, off);
+      const int64_t j = i - off[t];
+      g.p[t][j] = acc;
+      float v = acc;
+      if (buf.p[t]) {
+        v = mu * buf.p[t][j] + acc;
+        buf.p[t][j] = v;
+      }
+      prm.p[t][j] -= lr * v;
     }
-    prm.p[t][j] -= lr * v;
   }
   if (loss_part && blockIdx.x == 0)
     net_loss_finalize(loss_part, loss_out, nblk_fwd, seed_bump);
@@ -2003,7 +2046,7 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
   gp.p[4] = (float*)gwf1; gp.p[5] = (float*)gbf1;
   gp.p[6] = (float*)gwf2; gp.p[7] = (float*)gbf2;
   hipLaunchKernelGGL(net_gw_combine_kernel,
-                     dim3((GW_TOTAL + 255) / 256), dim3(256), 0,
+                     dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
                      S(stream), (const float*)part_ws, gp, nch,
                      (const float*)loss_part, (float*)loss_out,
                      grid_for(B, 1),
@@ -2066,7 +2109,7 @@ void net_fused_bwd_sgd(uintptr_t x, uintptr_t w2, uintptr_t wf1,
     bp.p[i] = (i < (int)buf_v.size()) ? (float*)buf_v[i] : nullptr;
   }
   hipLaunchKernelGGL(net_gw_combine_sgd_kernel,
-                     dim3((GW_TOTAL + 255) / 256), dim3(256), 0,
+                     dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
                      S(stream), (const float*)part_ws, gp, pp, bp, nch,
                      (float)lr, (float)mu, (const float*)loss_part,
                      (float*)loss_out, grid_for(B, 1),
