@@ -2,6 +2,7 @@
 loopback (the reference's own harness shape): every L2
 primitive the tutorial catalogues (tuto.md:77-202)."""
 
+import os
 import torch
 
 from dist_tuto_pth_amd import dist
@@ -183,3 +184,46 @@ def test_rank_world_backend():
 
 def test_ptp_demo_world2():
     launch(_fn_ptp_demo, 2)
+
+
+# ---------------------------------------------------------------------------
+# init methods beyond env:// (tuto.md:421-457: file:// with fcntl
+# locking, tcp://host:port)
+# ---------------------------------------------------------------------------
+def _fn_initmethod_worker(rank, size, init_method):
+    from dist_tuto_pth_amd import dist as d
+    d.init_process_group("gloo", init_method=init_method, world_size=size,
+                         rank=rank)
+    try:
+        t = torch.full((4,), float(rank + 1))
+        d.all_reduce(t, op=d.ReduceOp.SUM)
+        expect = float(sum(r + 1 for r in range(size)))
+        assert torch.allclose(t, torch.full((4,), expect))
+    finally:
+        d.destroy_process_group()
+
+
+def _spawn_initmethod(init_method, size=2):
+    import torch.multiprocessing as mp
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_fn_initmethod_worker,
+                         args=(r, size, init_method)) for r in range(size)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(120)
+    for r, p in enumerate(procs):
+        assert p.exitcode == 0, (r, p.exitcode)
+
+
+def test_init_method_file():
+    import tempfile
+    d = tempfile.mkdtemp()
+    path = os.path.join(d, "rdzv")
+    _spawn_initmethod(f"file://{path}")
+
+
+def test_init_method_tcp():
+    from dist_tuto_pth_amd.dist import _free_port
+    port = _free_port()
+    _spawn_initmethod(f"tcp://127.0.0.1:{port}")
