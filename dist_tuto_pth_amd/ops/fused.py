@@ -119,12 +119,12 @@ def net_fused_step(net, x: torch.Tensor, tgt: torch.Tensor) -> torch.Tensor:
     if "one" not in ws:
         ws["one"] = torch.ones((), device=x.device)
     if "loss_part" not in ws:
-        ws["loss_part"] = torch.empty(512, device=x.device)
+        ws["loss_part"] = torch.empty(4096, device=x.device)
     s = _stream()
-    # loss_part mode (B<=512): no prologue dispatch — the forward
-    # writes per-block loss partials; the combine kernel finalizes the
-    # loss scalar and bumps the dropout seed for the next step
-    lp = ws["loss_part"].data_ptr() if B <= 512 else 0
+    # loss_part mode: no prologue dispatch — the forward writes
+    # per-block loss partials (fwd grid <= 4096 at any B); the combine
+    # kernel finalizes the loss scalar and bumps the dropout seed
+    lp = ws["loss_part"].data_ptr()
     k.net_fused_fwd(
         x.data_ptr(), *[p.data_ptr() for p in params], tgt.data_ptr(),
         ws["p1"].data_ptr(), ws["idx1"].data_ptr(), ws["m2"].data_ptr(),
@@ -150,14 +150,9 @@ def net_fused_step_opt(net, x: torch.Tensor, tgt: torch.Tensor,
     """``net_fused_step`` with the SGD+momentum update fused into the
     combine kernel (one dispatch fewer; single-GPU training only — the
     DP path needs the all-reduce between combine and step).  ``opt``
-    must be a FusedSGD; its momentum buffers are updated in-kernel.
-    Falls back to step+opt.step() for B > 512."""
+    must be a FusedSGD; its momentum buffers are updated in-kernel."""
     k = load_native("_kernels")
     B = x.shape[0]
-    if B > 512:
-        loss = net_fused_step(net, x, tgt)
-        opt.step()
-        return loss
     ws = _ws(B, x.device)
     params = [net.conv1.weight, net.conv1.bias, net.conv2.weight,
               net.conv2.bias, net.fc1.weight, net.fc1.bias,
@@ -168,7 +163,7 @@ def net_fused_step_opt(net, x: torch.Tensor, tgt: torch.Tensor,
     if "one" not in ws:
         ws["one"] = torch.ones((), device=x.device)
     if "loss_part" not in ws:
-        ws["loss_part"] = torch.empty(512, device=x.device)
+        ws["loss_part"] = torch.empty(4096, device=x.device)
     s = _stream()
     lp = ws["loss_part"].data_ptr()
     k.net_fused_fwd(

@@ -124,6 +124,20 @@ def test_fused_step_opt_matches_separate_sgd_eval():
         assert torch.allclose(ba, bb, atol=1e-6)
 
 
+def test_fused_step_large_batch_matches_cpu():
+    """B > 512 now runs the same segmented-partial gw path (the
+    adaptive per-op path measured 4x slower at B=4096; profiles/)."""
+    net_c, net_g, x, tgt = _mk(11, B=1024)
+    flat = attach_flat_grads(net_g)
+    loss_g = net_fused_step(net_g, x.to(DEV), tgt.to(DEV)).clone()
+    torch.cuda.synchronize()
+    loss_c = F.nll_loss(net_c(x), tgt)
+    loss_c.backward()
+    assert torch.allclose(loss_g.cpu(), loss_c, atol=1e-5)
+    for (n, pc), pg in zip(net_c.named_parameters(), net_g.parameters()):
+        assert torch.allclose(pg.grad.cpu(), pc.grad, atol=5e-4), n
+
+
 def test_megakernel_matches_fused_eval():
     """Single-launch cooperative step == 6-dispatch fused path, exactly
     (same shared __device__ code, eval mode => deterministic)."""
