@@ -38,6 +38,9 @@ def main():
                    help="NHWC memory format (MIOpen fast path)")
     p.add_argument("--no-find", action="store_true",
                    help="disable MIOpen exhaustive kernel search")
+    p.add_argument("--dtype", default="fp32", choices=["fp32", "bf16"],
+                   help="bf16 = autocast compute (supplementary number; "
+                        "the headline config is fp32)")
     args = p.parse_args()
 
     # MIOpen exhaustive find: pick the fastest conv kernel per shape
@@ -75,14 +78,20 @@ def main():
     tgt = torch.randint(0, 1000, (args.batch,), generator=g).to(device)
     crit = torch.nn.CrossEntropyLoss()
 
+    import contextlib
+    amp = (torch.autocast("cuda", dtype=torch.bfloat16)
+           if args.dtype == "bf16" else contextlib.nullcontext())
+
     def step():
         opt.zero_grad(set_to_none=True)
         if ddp is not None:
-            loss = crit(ddp(x), tgt)
+            with amp:
+                loss = crit(ddp(x), tgt)
             loss.backward()
             ddp.finish_gradients()
         else:
-            loss = crit(model(x), tgt)
+            with amp:
+                loss = crit(model(x), tgt)
             loss.backward()
             if world > 1:
                 from dist_tuto_pth_amd.parallel import average_gradients
@@ -120,7 +129,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "fp32",
+            "dtype": args.dtype,
             "data": "synthetic",
             "config": {"model": "ResNet-50", "global_batch":
                        args.batch * world, "input": "3x224x224",

@@ -1,13 +1,17 @@
 """Fused whole-Net training step (GPU fast path).
 
-One kernel for the entire forward (conv1..log_softmax+NLL,
-train_dist.py:64-71 + :120) and one for the data backward, plus four
-chunked weight-gradient reductions — ~8 launches per step instead of
-~30 (the per-op pipeline is launch-bound at the reference's batch
-sizes; rocprof evidence in profiles/).  Numerically identical to the
-modular path in eval mode; dropout masks use the same device-seed
-stream but a different indexing, so train-mode losses match only in
-distribution.
+Four dispatches per training step instead of ~30 per-op launches:
+one kernel for the entire forward (conv1..log_softmax+NLL,
+train_dist.py:64-71 + :120, conv+pool register-fused, per-block loss
+partials), one for the data backward (sibling-workgroup split,
+4-wide register-blocked transposed conv), one segmented partial
+weight-gradient kernel ([conv2|fc1|conv1x8|fc2] tiles x batch chunks),
+and one combine kernel that also finalizes the loss, advances the
+dropout seed, and (single-GPU) applies the SGD+momentum update.
+rocprof evidence and the optimization ladder live in profiles/.
+Numerically identical to the modular path in eval mode; dropout masks
+use the same device-seed stream but a different indexing, so
+train-mode losses match only in distribution.
 """
 
 from __future__ import annotations
