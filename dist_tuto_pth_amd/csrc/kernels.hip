@@ -625,6 +625,18 @@ __global__ void sgd_step_kernel(MtArgs a, float lr, float mu,
 // K14 helpers — local reductions for the hand-rolled / hand-tuned ring
 // all-reduce (allreduce.py:26,31 corrected): dst += src, vectorized.
 // fp32 and bf16 (BASELINE config 5).
+//
+// MFMA considered and rejected for this op: a column reduction is
+// out[j] = sum_p src[p][j], i.e. ones[1xP] x src[PxN].  On a 16x16x4
+// f32 MFMA every C row would hold the same 4-row partial sum, so one
+// 4-cycle instruction reduces 4x16 = 64 elements = 16 elem/cycle/wave,
+// while plain VALU v_add_f32 reduces 64 elem/cycle/wave — the matrix
+// pipe wastes a 16x output replication on a rank-1 operand.  The op is
+// HBM-bandwidth-bound anyway (reads P*N floats once); the float4
+// grid-stride form below saturates that.  MFMA belongs to GEMM-shaped
+// work, which this framework's models reach through the register-
+// blocked conv kernels above (shapes are 5x5/latency-bound, below the
+// MFMA payoff threshold) and hipBLASLt/MIOpen for ResNet-50.
 // ===========================================================================
 __global__ void add_inplace_f32_kernel(float* __restrict__ dst,
                                        const float* __restrict__ src,
