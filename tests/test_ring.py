@@ -82,3 +82,19 @@ def test_chunked_average():
 
 def test_chunked_matches_builtin_allreduce():
     launch(_fn_vs_builtin, 3)
+
+
+def test_pad_chunks_cpu():
+    """_pad_chunks: 16-byte-aligned chunks, zero-fill, round-trip copy
+    (the alignment contract the xGMI exchange relies on)."""
+    from dist_tuto_pth_amd.algorithms.xgmi import _pad_chunks
+    for numel, size in ((64, 4), (37, 4), (5, 8), (1, 2), (257, 8)):
+        t = torch.arange(numel, dtype=torch.float32)
+        work, chunk, padded = _pad_chunks(t, size)
+        assert chunk * size >= numel
+        assert chunk % (16 // 4) == 0          # 16-byte alignment in fp32
+        assert torch.equal(work[:numel], t)
+        if padded:
+            assert work[numel:chunk * size].abs().sum() == 0
+        else:
+            assert work.data_ptr() == t.data_ptr()
