@@ -115,6 +115,9 @@ def main():
     results["gw conv1 tiles"] = time_fn(seg(T_CONV2 + T_FC1, T_CONV1))
     results["gw fc2 tiles"] = time_fn(seg(T_CONV2 + T_FC1 + T_CONV1, T_FC2))
     results["gw all tiles"] = time_fn(seg(0, T_CONV2 + T_FC1 + T_CONV1 + T_FC2))
+    results["gw combine"] = time_fn(lambda: k.net_gw_combine_raw(
+        ws["part"].data_ptr(), [p.grad.data_ptr() for p in params], nch,
+        s))
     results["sgd"] = time_fn(opt.step)
 
     for name, us in results.items():

@@ -2128,6 +2128,19 @@ void net_fused_bwd_sgd(uintptr_t x, uintptr_t w2, uintptr_t wf1,
                          ? (unsigned long long*)seed_dev : nullptr);
 }
 
+// raw combine launch (microbenchmarks: time the combine/sgd dispatch
+// in isolation)
+void net_gw_combine_raw(uintptr_t part_ws,
+                        const std::vector<uintptr_t>& grd_v, int nch,
+                        uintptr_t stream) {
+  GwPtrs gp{};
+  for (int i = 0; i < 8; ++i) gp.p[i] = (float*)grd_v[i];
+  hipLaunchKernelGGL(net_gw_combine_kernel,
+                     dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
+                     S(stream), (const float*)part_ws, gp, nch, nullptr,
+                     nullptr, 0, nullptr);
+}
+
 // raw tile-segment launch of the partial weight-gradient kernel
 // (microbenchmarks: time [conv2 | fc1 | conv1 | fc2] separately)
 void net_gw_partial_raw(uintptr_t x, uintptr_t p1_ws, uintptr_t p2_ws,
@@ -2301,6 +2314,7 @@ PYBIND11_MODULE(_kernels, m) {
   m.def("net_step", &net_step);
   m.def("net_step_available", &net_step_available);
   m.def("net_gw_partial_raw", &net_gw_partial_raw);
+  m.def("net_gw_combine_raw", &net_gw_combine_raw);
   m.def("net_fused_bwd_sgd", &net_fused_bwd_sgd);
   m.def("add_inplace", &add_inplace);
   m.def("reduce_columns", &reduce_columns);
