@@ -1,1 +1,2 @@
 from .net import Net  # noqa: F401
+from .mlp import MLP  # noqa: F401
