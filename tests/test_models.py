@@ -81,3 +81,29 @@ def test_mlp_trains_on_fixed_batch():
         opt.step()
         losses.append(float(loss))
     assert losses[-1] < losses[0] - 0.3, (losses[0], losses[-1])
+
+
+import pytest  # noqa: E402
+
+
+@pytest.mark.gpu
+def test_mlp_gpu_matches_cpu():
+    import torch
+    import torch.nn.functional as F
+    from dist_tuto_pth_amd.models import MLP
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    torch.manual_seed(5)
+    m = MLP((784, 200, 10), dropout=0.0).eval()
+    mg = MLP((784, 200, 10), dropout=0.0).eval().cuda()
+    mg.load_state_dict({k: v.cuda() for k, v in m.state_dict().items()})
+    x = torch.randn(16, 784)
+    tgt = torch.randint(0, 10, (16,))
+    loss_c = F.nll_loss(m(x), tgt)
+    loss_c.backward()
+    loss_g = F.nll_loss(mg(x.cuda()), tgt.cuda())
+    loss_g.backward()
+    torch.cuda.synchronize()
+    assert torch.allclose(loss_g.cpu(), loss_c, atol=1e-5)
+    for pc, pg in zip(m.parameters(), mg.parameters()):
+        assert torch.allclose(pg.grad.cpu(), pc.grad, atol=1e-4)
