@@ -62,13 +62,17 @@ __global__ void conv2d_fwd_kernel(const float* __restrict__ x,
                                   const float* __restrict__ bias,
                                   float* __restrict__ out,
                                   int B, int C, int H, int W,
-                                  int K, int R, int S_) {
+                                  int K, int R, int S_, int staged) {
   extern __shared__ __attribute__((aligned(16))) float smem[];
   const int OH = H - R + 1, OW = W - S_ + 1;
-  const int wn = K * C * R * S_;
-  float* ws = smem;  // [K*C*R*S] — shared by all threads, hot across K
-  for (int i = threadIdx.x; i < wn; i += blockDim.x) ws[i] = w[i];
-  __syncthreads();
+  // weights through LDS when they fit; global (L2-hot) otherwise
+  const float* ws = w;
+  if (staged) {
+    const int wn = K * C * R * S_;
+    for (int i = threadIdx.x; i < wn; i += blockDim.x) smem[i] = w[i];
+    __syncthreads();
+    ws = smem;
+  }
 
   // one thread per output element across the whole tensor: fills the
   // 256-CU chip at any batch size (the per-batch-block form left half
@@ -108,13 +112,16 @@ __global__ void conv2d_bwd_x_kernel(const float* __restrict__ gy,
                                     const float* __restrict__ w,
                                     float* __restrict__ gx,
                                     int B, int C, int H, int W,
-                                    int K, int R, int S_) {
+                                    int K, int R, int S_, int staged) {
   extern __shared__ __attribute__((aligned(16))) float smem[];
   const int OH = H - R + 1, OW = W - S_ + 1;
-  const int wn = K * C * R * S_;
-  float* ws = smem;
-  for (int i = threadIdx.x; i < wn; i += blockDim.x) ws[i] = w[i];
-  __syncthreads();
+  const float* ws = w;
+  if (staged) {
+    const int wn = K * C * R * S_;
+    for (int i = threadIdx.x; i < wn; i += blockDim.x) smem[i] = w[i];
+    __syncthreads();
+    ws = smem;
+  }
 
   const int64_t n_in = (int64_t)B * C * H * W;
   const int xn = C * H * W;
@@ -377,12 +384,18 @@ __global__ void linear_fwd_kernel(const float* __restrict__ x,
                                   const float* __restrict__ w,
                                   const float* __restrict__ bias,
                                   float* __restrict__ out,
-                                  int B, int K, int N, int fuse_relu) {
+                                  int B, int K, int N, int fuse_relu,
+                                  int staged) {
+  // weights staged through LDS when they fit (Net's fc shapes);
+  // read from global (L2-resident, shared by every block) otherwise
   extern __shared__ __attribute__((aligned(16))) float smem[];
-  float* ws = smem;  // [N*K]
-  const int wn = N * K;
-  for (int i = threadIdx.x; i < wn; i += blockDim.x) ws[i] = w[i];
-  __syncthreads();
+  const float* ws = w;
+  if (staged) {
+    const int wn = N * K;
+    for (int i = threadIdx.x; i < wn; i += blockDim.x) smem[i] = w[i];
+    __syncthreads();
+    ws = smem;
+  }
 
   const int64_t n_out = (int64_t)B * N;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -407,12 +420,15 @@ __global__ void linear_bwd_x_kernel(const float* __restrict__ gy,
                                     const float* __restrict__ out,
                                     const float* __restrict__ w,
                                     float* __restrict__ gx,
-                                    int B, int K, int N) {
+                                    int B, int K, int N, int staged) {
   extern __shared__ __attribute__((aligned(16))) float smem[];
-  float* ws = smem;  // [N*K]
-  const int wn = N * K;
-  for (int i = threadIdx.x; i < wn; i += blockDim.x) ws[i] = w[i];
-  __syncthreads();
+  const float* ws = w;
+  if (staged) {
+    const int wn = N * K;
+    for (int i = threadIdx.x; i < wn; i += blockDim.x) smem[i] = w[i];
+    __syncthreads();
+    ws = smem;
+  }
 
   const int64_t n_out = (int64_t)B * K;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -1698,12 +1714,14 @@ constexpr int BLK = 256;
 void conv2d_fwd(uintptr_t x, uintptr_t w, uintptr_t bias, uintptr_t out,
                 int B, int C, int H, int W, int K, int R, int S_,
                 uintptr_t stream) {
-  const int lds = K * C * R * S_ * sizeof(float);
+  const int64_t need = (int64_t)K * C * R * S_ * sizeof(float);
+  const int staged = need <= 64 * 1024 ? 1 : 0;
   const int OH = H - R + 1, OW = W - S_ + 1;
   const int64_t n = (int64_t)B * K * OH * OW;
   hipLaunchKernelGGL(conv2d_fwd_kernel, dim3(grid_for(n, BLK)), dim3(BLK),
-                     lds, S(stream), (const float*)x, (const float*)w,
-                     (const float*)bias, (float*)out, B, C, H, W, K, R, S_);
+                     staged ? (int)need : 0, S(stream), (const float*)x,
+                     (const float*)w, (const float*)bias, (float*)out, B,
+                     C, H, W, K, R, S_, staged);
 }
 
 void conv2d_bwd(uintptr_t x, uintptr_t w, uintptr_t gy, uintptr_t gx,
@@ -1711,11 +1729,13 @@ void conv2d_bwd(uintptr_t x, uintptr_t w, uintptr_t gy, uintptr_t gx,
                 int K, int R, int S_, uintptr_t stream) {
   const int OH = H - R + 1, OW = W - S_ + 1;
   {
-    const int lds = K * C * R * S_ * sizeof(float);
+    const int64_t need = (int64_t)K * C * R * S_ * sizeof(float);
+    const int staged = need <= 64 * 1024 ? 1 : 0;
     const int64_t n = (int64_t)B * C * H * W;
     hipLaunchKernelGGL(conv2d_bwd_x_kernel, dim3(grid_for(n, BLK)),
-                       dim3(BLK), lds, S(stream), (const float*)gy,
-                       (const float*)w, (float*)gx, B, C, H, W, K, R, S_);
+                       dim3(BLK), staged ? (int)need : 0, S(stream),
+                       (const float*)gy, (const float*)w, (float*)gx, B,
+                       C, H, W, K, R, S_, staged);
   }
   {
     const int wn = K * C * R * S_;
@@ -1811,23 +1831,27 @@ void dropout2d_bwd(uintptr_t gy, uintptr_t mask, uintptr_t gx,
 
 void linear_fwd(uintptr_t x, uintptr_t w, uintptr_t bias, uintptr_t out,
                 int B, int K, int N, bool fuse_relu, uintptr_t stream) {
-  const int lds = N * K * sizeof(float);
+  const int64_t need = (int64_t)N * K * sizeof(float);
+  const int staged = need <= 64 * 1024 ? 1 : 0;  // LDS budget per WG
   hipLaunchKernelGGL(linear_fwd_kernel,
-                     dim3(grid_for((int64_t)B * N, BLK)), dim3(BLK), lds,
+                     dim3(grid_for((int64_t)B * N, BLK)), dim3(BLK),
+                     staged ? (int)need : 0,
                      S(stream), (const float*)x, (const float*)w,
                      (const float*)bias, (float*)out, B, K, N,
-                     fuse_relu ? 1 : 0);
+                     fuse_relu ? 1 : 0, staged);
 }
 
 void linear_bwd(uintptr_t x, uintptr_t w, uintptr_t gy, uintptr_t out,
                 uintptr_t gx, uintptr_t gw, uintptr_t gb, int B, int K,
                 int N, uintptr_t stream) {
   {
-    const int lds = N * K * sizeof(float);
+    const int64_t need = (int64_t)N * K * sizeof(float);
+    const int staged = need <= 64 * 1024 ? 1 : 0;
     hipLaunchKernelGGL(linear_bwd_x_kernel,
-                       dim3(grid_for((int64_t)B * K, BLK)), dim3(BLK), lds,
+                       dim3(grid_for((int64_t)B * K, BLK)), dim3(BLK),
+                       staged ? (int)need : 0,
                        S(stream), (const float*)gy, (const float*)out,
-                       (const float*)w, (float*)gx, B, K, N);
+                       (const float*)w, (float*)gx, B, K, N, staged);
   }
   {
     const int wn = N * K;

@@ -222,3 +222,39 @@ def test_net_forward_backward_gpu_vs_cpu():
     for pc, pg in zip(net_c.parameters(), net_g.parameters()):
         assert torch.allclose(pg.grad.cpu(), pc.grad, atol=1e-3), \
             (pg.grad.cpu() - pc.grad).abs().max()
+
+
+def test_conv2d_large_weights_unstaged():
+    """Weights bigger than the 64 KB LDS budget take the global-read
+    path (staged=0) instead of failing the launch."""
+    torch.manual_seed(20)
+    x = torch.randn(4, 32, 16, 16, device=DEV, requires_grad=True)
+    w = torch.randn(64, 32, 5, 5, device=DEV, requires_grad=True)  # 204 KB
+    b = torch.randn(64, device=DEV, requires_grad=True)
+    out = ops.conv2d(x, w, b)
+    xc = x.detach().cpu().requires_grad_(True)
+    wc = w.detach().cpu().requires_grad_(True)
+    bc = b.detach().cpu().requires_grad_(True)
+    ref = F.conv2d(xc, wc, bc)
+    assert torch.allclose(out.cpu(), ref, atol=2e-3, rtol=1e-3)
+    out.sum().backward()
+    ref.sum().backward()
+    assert torch.allclose(x.grad.cpu(), xc.grad, atol=2e-3, rtol=1e-3)
+    assert torch.allclose(w.grad.cpu(), wc.grad, atol=2e-2, rtol=1e-3)
+
+
+def test_linear_large_weights_unstaged():
+    torch.manual_seed(21)
+    x = torch.randn(16, 784, device=DEV, requires_grad=True)
+    w = torch.randn(200, 784, device=DEV, requires_grad=True)  # 627 KB
+    b = torch.randn(200, device=DEV, requires_grad=True)
+    out = ops.linear(x, w, b)
+    xc = x.detach().cpu().requires_grad_(True)
+    wc = w.detach().cpu().requires_grad_(True)
+    bc = b.detach().cpu().requires_grad_(True)
+    ref = F.linear(xc, wc, bc)
+    assert torch.allclose(out.cpu(), ref, atol=2e-3, rtol=1e-3)
+    out.sum().backward()
+    ref.sum().backward()
+    assert torch.allclose(x.grad.cpu(), xc.grad, atol=2e-3, rtol=1e-3)
+    assert torch.allclose(w.grad.cpu(), wc.grad, atol=2e-2, rtol=1e-3)
