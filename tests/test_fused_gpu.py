@@ -224,3 +224,17 @@ def test_fused_training_convergence():
     first = sum(losses[:10]) / 10
     last = sum(losses[-10:]) / 10
     assert last < first - 0.12, (first, last)
+
+
+def test_fused_step_tiny_odd_batch():
+    """B=3: odd, smaller than every tile/chunk constant — exercises the
+    split/bchunk/nch edge paths."""
+    net_c, net_g, x, tgt = _mk(12, B=3)
+    attach_flat_grads(net_g)
+    loss_g = net_fused_step(net_g, x.to(DEV), tgt.to(DEV)).clone()
+    torch.cuda.synchronize()
+    loss_c = F.nll_loss(net_c(x), tgt)
+    loss_c.backward()
+    assert torch.allclose(loss_g.cpu(), loss_c, atol=1e-5)
+    for (n, pc), pg in zip(net_c.named_parameters(), net_g.parameters()):
+        assert torch.allclose(pg.grad.cpu(), pc.grad, atol=2e-4), n
