@@ -61,14 +61,26 @@ def main():
     rank = int(os.environ.get("RANK", 0))
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
 
-    dev_idx = local_rank % torch.cuda.device_count()
-    if world > 1:
-        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        os.environ.setdefault("MASTER_PORT", "29500")
-        dist.init_process_group("rccl", world_size=world, rank=rank,
-                                device_id=dev_idx)
-    torch.cuda.set_device(dev_idx)
-    device = f"cuda:{dev_idx}"
+    # DTP_BENCH_CPU=1: run the IDENTICAL launch path (torchrun env
+    # rendezvous, broadcast, timed loop, JSON line) on CPU/gloo so the
+    # driver contract is testable without a GPU.  Never auto-selected.
+    cpu_mode = os.environ.get("DTP_BENCH_CPU") == "1"
+    if cpu_mode:
+        if world > 1:
+            os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+            os.environ.setdefault("MASTER_PORT", "29500")
+            dist.init_process_group("gloo", world_size=world, rank=rank)
+        device = "cpu"
+        args.no_fused = True
+    else:
+        dev_idx = local_rank % torch.cuda.device_count()
+        if world > 1:
+            os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+            os.environ.setdefault("MASTER_PORT", "29500")
+            dist.init_process_group("rccl", world_size=world, rank=rank,
+                                    device_id=dev_idx)
+        torch.cuda.set_device(dev_idx)
+        device = f"cuda:{dev_idx}"
 
     torch.manual_seed(1234)
     model = Net().to(device)
@@ -134,7 +146,8 @@ def main():
     def barrier_sync():
         if world > 1:
             dist.barrier()
-        torch.cuda.synchronize()
+        if not cpu_mode:
+            torch.cuda.synchronize()
 
     graph = None
     if args.graph:

@@ -1,0 +1,63 @@
+"""The driver contract of bench.py, exercised end-to-end on CPU
+(DTP_BENCH_CPU=1): the exact torchrun launch line the benchmark driver
+uses, env rendezvous, parameter broadcast, timed loop, and the single
+JSON result line with all required fields."""
+
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+REQUIRED = ["metric", "value", "unit", "n_gpus", "steps", "warmup",
+            "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+            "dtype", "data", "config"]
+
+
+def _run_bench(nproc, extra=()):
+    from dist_tuto_pth_amd.dist import _free_port
+    env = dict(os.environ)
+    env["DTP_BENCH_CPU"] = "1"
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           f"--nproc-per-node={nproc}", "--master-addr", "127.0.0.1",
+           "--master-port", str(_free_port()), "bench.py",
+           "--gpus", str(nproc), "--steps", "3", "--warmup", "1",
+           "--batch", "8", *extra]
+    out = subprocess.run(cmd, cwd=ROOT, env=env, capture_output=True,
+                         text=True, timeout=240)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [ln for ln in out.stdout.splitlines() if ln.startswith("{")]
+    assert len(lines) == 1, out.stdout  # exactly ONE JSON line (rank 0)
+    return json.loads(lines[0])
+
+
+@pytest.mark.timeout(300)
+def test_bench_torchrun_world2_json_contract():
+    d = _run_bench(2)
+    for k in REQUIRED:
+        assert k in d, k
+    assert d["n_gpus"] == 2
+    assert d["steps"] == 3 and d["warmup"] == 1
+    assert d["metric"] == "ConvNet samples/sec"
+    assert d["higher_is_better"] is True
+    assert d["scaling"] == "weak"
+    assert d["data"] == "synthetic"
+    assert d["config"]["global_batch"] == 16        # 8 per rank, weak
+    assert d["config"]["parallelism"] == "dp2"
+    assert d["value"] > 0 and d["ms_per_step"] > 0
+
+
+def test_bench_single_process_json():
+    env = dict(os.environ)
+    env["DTP_BENCH_CPU"] = "1"
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "2", "--warmup", "1",
+         "--batch", "8"],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=180)
+    assert out.returncode == 0, out.stderr[-2000:]
+    d = json.loads([ln for ln in out.stdout.splitlines()
+                    if ln.startswith("{")][-1])
+    assert d["n_gpus"] == 1
