@@ -118,6 +118,13 @@ def main():
     results["gw combine"] = time_fn(lambda: k.net_gw_combine_raw(
         ws["part"].data_ptr(), [p.grad.data_ptr() for p in params], nch,
         s))
+    ws.setdefault("loss_part", torch.empty(4096, device=dev))
+    bufs = [b.data_ptr() for b in opt._bufs]
+    results["gw combine+sgd+loss"] = time_fn(
+        lambda: k.net_gw_combine_sgd_raw(
+            ws["part"].data_ptr(), [p.grad.data_ptr() for p in params],
+            pp, bufs, nch, 0.01, 0.5, ws["loss_part"].data_ptr(),
+            ws["loss"].data_ptr(), 128, s))
     results["sgd"] = time_fn(opt.step)
 
     for name, us in results.items():

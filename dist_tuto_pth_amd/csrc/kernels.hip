@@ -2165,6 +2165,27 @@ void net_gw_combine_raw(uintptr_t part_ws,
                      nullptr, 0, nullptr);
 }
 
+// raw combine+sgd launch (microbenchmarks)
+void net_gw_combine_sgd_raw(uintptr_t part_ws,
+                            const std::vector<uintptr_t>& grd_v,
+                            const std::vector<uintptr_t>& prm_v,
+                            const std::vector<uintptr_t>& buf_v,
+                            int nch, double lr, double mu,
+                            uintptr_t loss_part, uintptr_t loss_out,
+                            int nblk_fwd, uintptr_t stream) {
+  GwPtrs gp{}, pp{}, bp{};
+  for (int i = 0; i < 8; ++i) {
+    gp.p[i] = (float*)grd_v[i];
+    pp.p[i] = (float*)prm_v[i];
+    bp.p[i] = (i < (int)buf_v.size()) ? (float*)buf_v[i] : nullptr;
+  }
+  hipLaunchKernelGGL(net_gw_combine_sgd_kernel,
+                     dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
+                     S(stream), (const float*)part_ws, gp, pp, bp, nch,
+                     (float)lr, (float)mu, (const float*)loss_part,
+                     (float*)loss_out, nblk_fwd, nullptr);
+}
+
 // raw tile-segment launch of the partial weight-gradient kernel
 // (microbenchmarks: time [conv2 | fc1 | conv1 | fc2] separately)
 void net_gw_partial_raw(uintptr_t x, uintptr_t p1_ws, uintptr_t p2_ws,
@@ -2339,6 +2360,7 @@ PYBIND11_MODULE(_kernels, m) {
   m.def("net_step_available", &net_step_available);
   m.def("net_gw_partial_raw", &net_gw_partial_raw);
   m.def("net_gw_combine_raw", &net_gw_combine_raw);
+  m.def("net_gw_combine_sgd_raw", &net_gw_combine_sgd_raw);
   m.def("net_fused_bwd_sgd", &net_fused_bwd_sgd);
   m.def("add_inplace", &add_inplace);
   m.def("reduce_columns", &reduce_columns);
