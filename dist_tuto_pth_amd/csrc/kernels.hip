@@ -2202,6 +2202,30 @@ void net_gw_partial_raw(uintptr_t x, uintptr_t p1_ws, uintptr_t p2_ws,
                      (float*)part_ws, B, bchunk, tile_base);
 }
 
+// cooperative grid-barrier cost probe: `nsync` grid.sync()s and
+// nothing else, at fwd-kernel-like occupancy (32 KB LDS, 256 thr).
+// Times the lever that decides whether mid-kernel sibling exchanges
+// (fwd split, SURVEY-future) can beat kernel boundaries.
+__global__ void __launch_bounds__(256)
+barrier_probe_kernel(int nsync, float* sink) {
+  __shared__ float smem[8192];
+  smem[threadIdx.x] = (float)threadIdx.x;
+  cg::grid_group grid = cg::this_grid();
+  for (int i = 0; i < nsync; ++i) grid.sync();
+  if (threadIdx.x == 0 && blockIdx.x == 0) sink[0] = smem[0];
+}
+
+void barrier_probe(int nblk, int nsync, uintptr_t sink, uintptr_t stream) {
+  int dev = 0;
+  HIP_CHECK(hipGetDevice(&dev));
+  int nsync_v = nsync;
+  float* sinkp = (float*)sink;
+  void* args[] = {&nsync_v, &sinkp};
+  HIP_CHECK(hipLaunchCooperativeKernel(
+      reinterpret_cast<const void*>(barrier_probe_kernel), dim3(nblk),
+      dim3(256), args, 0, S(stream)));
+}
+
 // ---- single-launch training step (cooperative) --------------------------
 int net_step_max_blocks() {
   static int cached = -2;
@@ -2361,6 +2385,7 @@ PYBIND11_MODULE(_kernels, m) {
   m.def("net_gw_partial_raw", &net_gw_partial_raw);
   m.def("net_gw_combine_raw", &net_gw_combine_raw);
   m.def("net_gw_combine_sgd_raw", &net_gw_combine_sgd_raw);
+  m.def("barrier_probe", &barrier_probe);
   m.def("net_fused_bwd_sgd", &net_fused_bwd_sgd);
   m.def("add_inplace", &add_inplace);
   m.def("reduce_columns", &reduce_columns);
