@@ -2226,6 +2226,52 @@ void barrier_probe(int nblk, int nsync, uintptr_t sink, uintptr_t stream) {
       dim3(256), args, 0, S(stream)));
 }
 
+// hand-rolled resident-grid flag barrier probe (the guide's G16
+// last-arriver pattern: agent-scope acq/rel on a generation counter,
+// relaxed poll).  Launched cooperatively ONLY for the co-residency
+// guarantee; measures what a grid barrier costs without
+// cooperative-groups' sync machinery (17-42 us, see barrier_probe).
+__global__ void __launch_bounds__(256)
+flag_barrier_probe_kernel(int nsync, unsigned int* bar, float* sink) {
+  __shared__ float smem[8192];
+  smem[threadIdx.x] = (float)threadIdx.x;
+  for (int i = 0; i < nsync; ++i) {
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      const unsigned int gen = __hip_atomic_load(
+          &bar[1], __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_AGENT);
+      const unsigned int arrived =
+          __hip_atomic_fetch_add(&bar[0], 1u, __ATOMIC_ACQ_REL,
+                                 __HIP_MEMORY_SCOPE_AGENT) + 1;
+      if (arrived == gridDim.x) {
+        __hip_atomic_store(&bar[0], 0u, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+        __hip_atomic_fetch_add(&bar[1], 1u, __ATOMIC_RELEASE,
+                               __HIP_MEMORY_SCOPE_AGENT);
+      } else {
+        while (__hip_atomic_load(&bar[1], __ATOMIC_RELAXED,
+                                 __HIP_MEMORY_SCOPE_AGENT) == gen)
+          __builtin_amdgcn_s_sleep(8);
+        (void)__hip_atomic_load(&bar[1], __ATOMIC_ACQUIRE,
+                                __HIP_MEMORY_SCOPE_AGENT);
+      }
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0 && blockIdx.x == 0) sink[0] = smem[0];
+}
+
+void flag_barrier_probe(int nblk, int nsync, uintptr_t bar,
+                        uintptr_t sink, uintptr_t stream) {
+  int nsync_v = nsync;
+  unsigned int* barp = (unsigned int*)bar;
+  float* sinkp = (float*)sink;
+  void* args[] = {&nsync_v, &barp, &sinkp};
+  HIP_CHECK(hipLaunchCooperativeKernel(
+      reinterpret_cast<const void*>(flag_barrier_probe_kernel),
+      dim3(nblk), dim3(256), args, 0, S(stream)));
+}
+
 // ---- single-launch training step (cooperative) --------------------------
 int net_step_max_blocks() {
   static int cached = -2;
@@ -2386,6 +2432,7 @@ PYBIND11_MODULE(_kernels, m) {
   m.def("net_gw_combine_raw", &net_gw_combine_raw);
   m.def("net_gw_combine_sgd_raw", &net_gw_combine_sgd_raw);
   m.def("barrier_probe", &barrier_probe);
+  m.def("flag_barrier_probe", &flag_barrier_probe);
   m.def("net_fused_bwd_sgd", &net_fused_bwd_sgd);
   m.def("add_inplace", &add_inplace);
   m.def("reduce_columns", &reduce_columns);
