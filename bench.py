@@ -73,7 +73,13 @@ def main():
         device = "cpu"
         args.no_fused = True
     else:
-        dev_idx = local_rank % torch.cuda.device_count()
+        n_dev = torch.cuda.device_count()
+        if world > n_dev:
+            raise RuntimeError(
+                f"world_size {world} exceeds visible GPUs ({n_dev}); "
+                f"one rank maps to one MI355X (RCCL does not support "
+                f"multiple ranks per device)")
+        dev_idx = local_rank % n_dev
         if world > 1:
             os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
             os.environ.setdefault("MASTER_PORT", "29500")

@@ -65,8 +65,9 @@ def main():
     p.add_argument("--iters", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--algos", default="rccl,fullmesh,ring")
-    p.add_argument("--check", action="store_true",
-                   help="verify results against RCCL first")
+    p.add_argument("--no-check", action="store_true",
+                   help="skip verifying the hand-rolled algorithms "
+                        "against RCCL before timing")
     args = p.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", 1))
@@ -89,7 +90,7 @@ def main():
     g = torch.Generator().manual_seed(42 + rank)
     base = torch.randn(numel, generator=g).to(dtype).cuda()
 
-    if args.check and world > 1:
+    if not args.no_check and world > 1:
         ref = base.clone()
         dist.all_reduce(ref, op=dist.ReduceOp.SUM)
         for algo in ("fullmesh", "ring"):
