@@ -48,13 +48,9 @@ def build(force: bool = False):
         # (output, sources, extra flags)
         ("_rcclx.so", [os.path.join(CSRC, "rcclx.cpp")],
          ["-L/opt/rocm/lib", "-lrccl"]),
-        ("_kernels.so", [os.path.join(CSRC, "kernels.hip"),
-                         os.path.join(CSRC, "kernels_binding.cpp")], []),
+        ("_kernels.so", [os.path.join(CSRC, "kernels.hip")], []),
     ]
     for out_name, srcs, extra in targets:
-        srcs = [s for s in srcs if os.path.exists(s)]
-        if not srcs:
-            continue
         out = os.path.join(OUT, out_name)
         if force or _needs_build(out, srcs):
             _run(common + srcs + extra + ["-o", out])

@@ -1242,7 +1242,9 @@ net_fused_bwd_kernel(
 #define OFF_BF1 21280     // 50
 #define OFF_WF2 21330     // 500
 #define OFF_BF2 21830     // 10
-#define T_CONV2 20        // ceil(5020/256)
+#define T_CONV2 20        // one block per conv2 output channel
+static_assert(T_CONV2 == N_C2K, "conv2 gw tiling is one block per "
+              "output channel");
 #define T_FC1 63          // ceil(16050/256)
 #define T_CONV1 8         // 8 sub-blocks over 3-output-row bands
 #define T_FC2 2           // ceil(510/256)
@@ -1490,24 +1492,27 @@ __device__ __forceinline__ int net_gw_tensor_of(int i, const int* off) {
 // finalize the per-block loss partials written by the forward kernel
 // (loss_part mode) and advance the dropout seed for the NEXT step —
 // runs in block 0 of a combine kernel, replacing the step-prologue
-// dispatch (~4.5 us device floor) entirely.
+// dispatch (~4.5 us device floor) entirely.  The seed convention is
+// bump-AFTER everywhere (both the loss_part and the legacy autograd
+// path): the step consumes the current seed, the combine advances it.
 __device__ __forceinline__ void net_loss_finalize(
     const float* __restrict__ loss_part, float* __restrict__ loss_out,
     int nblk_fwd, unsigned long long* seed_bump) {
-  __shared__ float red[256];
-  float v = 0.f;
-  for (int i = threadIdx.x; i < nblk_fwd; i += 256) v += loss_part[i];
-  red[threadIdx.x] = v;
-  __syncthreads();
-  #pragma unroll
-  for (int s2 = 128; s2 > 0; s2 >>= 1) {
-    if (threadIdx.x < s2) red[threadIdx.x] += red[threadIdx.x + s2];
+  if (loss_part) {
+    __shared__ float red[256];
+    float v = 0.f;
+    for (int i = threadIdx.x; i < nblk_fwd; i += 256) v += loss_part[i];
+    red[threadIdx.x] = v;
     __syncthreads();
+    #pragma unroll
+    for (int s2 = 128; s2 > 0; s2 >>= 1) {
+      if (threadIdx.x < s2) red[threadIdx.x] += red[threadIdx.x + s2];
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) *loss_out = red[0];
   }
-  if (threadIdx.x == 0) {
-    *loss_out = red[0];
-    if (seed_bump) *seed_bump += 0x9E3779B97F4A7C15ull;
-  }
+  if (threadIdx.x == 0 && seed_bump)
+    *seed_bump += 0x9E3779B97F4A7C15ull;
 }
 
 __global__ void net_gw_combine_kernel(const float* __restrict__ part,
@@ -1528,7 +1533,7 @@ __global__ void net_gw_combine_kernel(const float* __restrict__ part,
       g.p[t][i - off[t]] = acc;
     }
   }
-  if (loss_part && blockIdx.x == 0)
+  if ((loss_part || seed_bump) && blockIdx.x == 0)
     net_loss_finalize(loss_part, loss_out, nblk_fwd, seed_bump);
 }
 
@@ -1564,7 +1569,7 @@ __global__ void net_gw_combine_sgd_kernel(const float* __restrict__ part,
       prm.p[t][j] -= lr * v;
     }
   }
-  if (loss_part && blockIdx.x == 0)
+  if ((loss_part || seed_bump) && blockIdx.x == 0)
     net_loss_finalize(loss_part, loss_out, nblk_fwd, seed_bump);
 }
 
@@ -1956,13 +1961,14 @@ void net_fused_fwd(uintptr_t x, uintptr_t w1, uintptr_t b1, uintptr_t w2,
                    uintptr_t seed_dev, int B,
                    bool training, uintptr_t stream) {
   // legacy mode (loss_part == 0): prologue dispatch zeroes the loss
-  // scalar and bumps the seed, fwd accumulates the loss atomically.
-  // loss_part mode: no prologue — per-block partials, finalized (and
-  // seed bumped) by the combine kernel of the same step.
+  // scalar, fwd accumulates the loss atomically.  loss_part mode: no
+  // prologue — per-block partials, finalized by the combine kernel.
+  // Both modes bump the dropout seed AFTER the step, in the combine
+  // kernel of net_fused_bwd (one convention, so the autograd path and
+  // the fused-step path can interleave without mask reuse).
   if (!loss_part)
     hipLaunchKernelGGL(step_prologue_kernel, dim3(1), dim3(64), 0,
-                       S(stream),
-                       training ? (unsigned long long*)seed_dev : nullptr,
+                       S(stream), (unsigned long long*)nullptr,
                        (float*)loss);
   hipLaunchKernelGGL(net_fused_fwd_kernel, dim3(grid_for(B, 1)), dim3(256),
                      0, S(stream), (const float*)x, (const float*)w1,
@@ -2088,13 +2094,13 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
                      S(stream), (const float*)part_ws, gp, nch,
                      (const float*)loss_part, (float*)loss_out,
                      grid_for(B, 1),
-                     (loss_part && training)
+                     (training && seed_dev)
                          ? (unsigned long long*)seed_dev : nullptr);
 }
 
 // net_fused_bwd + the optimizer update fused into the combine kernel
-// (single-GPU path: no all-reduce between combine and step).  Falls
-// back is the caller's job for B > 512 (throws here).
+// (single-GPU path: no all-reduce between combine and step).  The
+// segmented-partial weight-gradient path handles any batch size.
 void net_fused_bwd_sgd(uintptr_t x, uintptr_t w2, uintptr_t wf1,
                        uintptr_t wf2, uintptr_t tgt, uintptr_t gl,
                        uintptr_t p1_ws, uintptr_t idx1_ws,
@@ -2148,7 +2154,7 @@ void net_fused_bwd_sgd(uintptr_t x, uintptr_t w2, uintptr_t wf1,
                      S(stream), (const float*)part_ws, gp, pp, bp, nch,
                      (float)lr, (float)mu, (const float*)loss_part,
                      (float*)loss_out, grid_for(B, 1),
-                     (loss_part && training)
+                     (training && seed_dev)
                          ? (unsigned long long*)seed_dev : nullptr);
 }
 

@@ -12,6 +12,13 @@ rocprof evidence and the optimization ladder live in profiles/.
 Numerically identical to the modular path in eval mode; dropout masks
 use the same device-seed stream but a different indexing, so
 train-mode losses match only in distribution.
+
+Seed convention (all fused paths): a training step CONSUMES the current
+device seed and the combine kernel at the end of backward ADVANCES it —
+bump-after everywhere, so the autograd path (net_fused_loss) and the
+fused-step paths can interleave without reusing dropout masks.  A
+train-mode forward without a matching backward does not advance the
+seed.
 """
 
 from __future__ import annotations
@@ -41,6 +48,11 @@ def _ws(B: int, device) -> Dict[str, torch.Tensor]:
             # partial rows are GW_ROW = 21840 + 7*260 wide (conv1's 8
             # weight-grad sub-blocks write disjoint slices; kernels.hip)
             "ga1": f(B, 5760), "part": f(32, 23660), "loss": f(()),
+            # loss_part must cover the LARGEST forward grid any caller
+            # can launch (grid_for(B,1) <= 4096) — sized once here so
+            # every entry point shares a safe buffer regardless of
+            # which ran first (advisor r1 finding #1)
+            "loss_part": f(4096), "one": torch.ones((), device=device),
         }
         _ws_cache[key] = w
     return w
@@ -101,8 +113,8 @@ class _NetFusedLoss(torch.autograd.Function):
             ws["glog"].data_ptr(), ws["gh1"].data_ptr(),
             ws["ga2"].data_ptr(), ws["ga1"].data_ptr(),
             ws["part"].data_ptr(),
-            *[g.data_ptr() for g in grads], B, training, 0, 0, 0,
-            _stream())
+            *[g.data_ptr() for g in grads], B, training, 0, 0,
+            _seed_ptr(x.device), _stream())
         return (None, *grads, None, None)
 
 
@@ -120,10 +132,6 @@ def net_fused_step(net, x: torch.Tensor, tgt: torch.Tensor) -> torch.Tensor:
     for p in params:
         if p.grad is None:
             p.grad = torch.empty_like(p)
-    if "one" not in ws:
-        ws["one"] = torch.ones((), device=x.device)
-    if "loss_part" not in ws:
-        ws["loss_part"] = torch.empty(4096, device=x.device)
     s = _stream()
     # loss_part mode: no prologue dispatch — the forward writes
     # per-block loss partials (fwd grid <= 4096 at any B); the combine
@@ -164,10 +172,6 @@ def net_fused_step_opt(net, x: torch.Tensor, tgt: torch.Tensor,
     for p in params:
         if p.grad is None:
             p.grad = torch.empty_like(p)
-    if "one" not in ws:
-        ws["one"] = torch.ones((), device=x.device)
-    if "loss_part" not in ws:
-        ws["loss_part"] = torch.empty(4096, device=x.device)
     s = _stream()
     lp = ws["loss_part"].data_ptr()
     k.net_fused_fwd(
@@ -224,10 +228,6 @@ def net_fused_train_step(net, x: torch.Tensor, tgt: torch.Tensor,
     for p in params:
         if p.grad is None:
             p.grad = torch.empty_like(p)
-    if "one" not in ws:
-        ws["one"] = torch.ones((), device=x.device)
-    if "loss_part" not in ws:
-        ws["loss_part"] = torch.empty(512, device=x.device)
     if do_sgd:
         lr, mu = opt.lr, opt.momentum
         bufs = [b.data_ptr() for b in opt._bufs] if opt._bufs else []
