@@ -44,16 +44,18 @@ def ring_all_reduce(send: torch.Tensor, recv: torch.Tensor, g=None):
     right = (rank + 1) % size
 
     for i in range(size - 1):
+        # dist.sendrecv is the deadlock-free form of the reference's
+        # isend(right) + recv(left) + wait() step (allreduce.py:24-32):
+        # on RCCL both transfers are posted inside one group so the
+        # send/recv cycle around the ring cannot stall (see
+        # dist.sendrecv docstring); the i%2 buffer swap still forwards
+        # last step's received buffer without a copy.
         if i % 2 == 0:
-            # send send_buff, receive into recv_buff
-            send_req = dist.isend(send_buff, right, g)
-            dist.recv(recv_buff, left, g)
+            dist.sendrecv(send_buff, right, recv_buff, left, g)
             accum += recv_buff
         else:
-            send_req = dist.isend(recv_buff, right, g)
-            dist.recv(send_buff, left, g)
+            dist.sendrecv(recv_buff, right, send_buff, left, g)
             accum += send_buff
-        send_req.wait()
     recv.copy_(accum)
 
 
@@ -92,10 +94,8 @@ def chunked_ring_all_reduce(tensor: torch.Tensor, g=None,
     for i in range(size - 1):
         send_idx = (rank - i) % size
         recv_idx = (rank - i - 1) % size
-        req = dist.isend(chunks[send_idx], right, g)
-        dist.recv(tmp, left, g)
+        dist.sendrecv(chunks[send_idx], right, tmp, left, g)
         chunks[recv_idx] += tmp
-        req.wait()
 
     owned = (rank + 1) % size
     if average:
@@ -105,9 +105,7 @@ def chunked_ring_all_reduce(tensor: torch.Tensor, g=None,
     for i in range(size - 1):
         send_idx = (owned - i) % size
         recv_idx = (owned - i - 1) % size
-        req = dist.isend(chunks[send_idx], right, g)
-        dist.recv(chunks[recv_idx], left, g)
-        req.wait()
+        dist.sendrecv(chunks[send_idx], right, chunks[recv_idx], left, g)
 
     if padded.data_ptr() != flat.data_ptr():
         flat.copy_(padded[:n])

@@ -50,6 +50,7 @@ __all__ = [
     "recv",
     "isend",
     "irecv",
+    "sendrecv",
     "broadcast",
     "reduce",
     "all_reduce",
@@ -328,6 +329,24 @@ def irecv(tensor: torch.Tensor, src: Optional[int] = None, g=None, tag: int = 0)
     return gr._impl.recv(tensor, src, blocking=False)
 
 
+def sendrecv(send_tensor: torch.Tensor, dst: int,
+             recv_tensor: torch.Tensor, src: int, g=None):
+    """Paired exchange: send ``send_tensor`` to ``dst`` while receiving
+    ``recv_tensor`` from ``src``, completing both before returning.
+
+    This is the safe form of the reference's ring step
+    (``isend(right)`` + blocking ``recv(left)``, allreduce.py:24-25).
+    On RCCL the un-paired pattern is a latent deadlock: every rank's
+    send kernel sits on the stream ahead of its recv kernel, and each
+    send waits for the *peer's* recv — a cycle once the message exceeds
+    RCCL's internal buffering.  Here both ops are posted inside one
+    ``ncclGroupStart``/``End`` so RCCL fuses them into a single
+    deadlock-free kernel; on gloo (CPU sockets, buffered) it is the
+    plain isend+recv pair."""
+    gr = _resolve(g)
+    return gr._impl.sendrecv(send_tensor, dst, recv_tensor, src)
+
+
 # --------------------------------------------------------------------------
 # the six collectives (tuto.md:197-202)
 # --------------------------------------------------------------------------
@@ -466,6 +485,13 @@ class _GlooBackend:
             return self._tdist.recv(tensor, d, group=self._group)
         req = self._tdist.irecv(tensor, d, group=self._group)
         return Work(req.wait)
+
+    def sendrecv(self, send_tensor, dst, recv_tensor, src):
+        # gloo's socket transport buffers sends, so the plain pair is
+        # already deadlock-free here
+        req = self.send(send_tensor, dst, blocking=False)
+        self.recv(recv_tensor, src, blocking=True)
+        req.wait()
 
     # -- collectives ------------------------------------------------------
     def _gr(self, rank_in_group):

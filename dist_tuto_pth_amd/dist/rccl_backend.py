@@ -155,6 +155,23 @@ class _RcclBackend:
         ev = self._rx.record_event(_stream())
         return Work(lambda: self._rx.event_wait(ev))
 
+    def sendrecv(self, send_t, dst, recv_t, src):
+        """Paired exchange in ONE RCCL group — the deadlock-free form
+        of the ring step (see dist.sendrecv).  Both transfers are fused
+        into a single kernel on the current stream; the call returns
+        when they are enqueued (stream-ordered consumers need no
+        wait)."""
+        _check(send_t)
+        _check(recv_t)
+        stream = _stream()
+        self._comm.group_start()
+        self._comm.send(send_t.data_ptr(), send_t.numel(),
+                        _DTYPE[send_t.dtype], dst, stream)
+        self._comm.recv(recv_t.data_ptr(), recv_t.numel(),
+                        _DTYPE[recv_t.dtype], src, stream)
+        self._comm.group_end()
+        torch.cuda.current_stream().synchronize()
+
     # ------------------------------------------------------------------
     # collectives — enqueued on the caller's current stream; Work.wait()
     # is a no-op for stream-ordered consumers (matching torch.distributed

@@ -87,10 +87,10 @@ def test_chunked_matches_builtin_allreduce():
 def test_pad_chunks_cpu():
     """_pad_chunks: 16-byte-aligned chunks, zero-fill, round-trip copy
     (the alignment contract the xGMI exchange relies on)."""
-    from dist_tuto_pth_amd.algorithms.xgmi import _pad_chunks
+    from dist_tuto_pth_amd.algorithms.xgmi import _buf, _pad_chunks
     for numel, size in ((64, 4), (37, 4), (5, 8), (1, 2), (257, 8)):
         t = torch.arange(numel, dtype=torch.float32)
-        work, chunk, padded = _pad_chunks(t, size)
+        work, chunk, padded = _pad_chunks(t, size, _buf)
         assert chunk * size >= numel
         assert chunk % (16 // 4) == 0          # 16-byte alignment in fp32
         assert torch.equal(work[:numel], t)
