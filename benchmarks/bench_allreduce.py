@@ -28,12 +28,12 @@ from dist_tuto_pth_amd import dist  # noqa: E402
 from dist_tuto_pth_amd.algorithms.xgmi import xgmi_all_reduce  # noqa: E402
 
 
-def run_algo(algo, t, iters, warmup, world):
+def run_algo(algo, t, iters, warmup, world, depth=4):
     def op():
         if algo == "rccl":
             dist.all_reduce(t, op=dist.ReduceOp.SUM)
         else:
-            xgmi_all_reduce(t, algo=algo)
+            xgmi_all_reduce(t, algo=algo, depth=depth)
 
     for _ in range(warmup):
         op()
@@ -65,6 +65,10 @@ def main():
     p.add_argument("--iters", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--algos", default="rccl,fullmesh,ring")
+    p.add_argument("--pipeline-depth", type=int, default=4,
+                   help="fullmesh sub-chunks per owned chunk: exchange "
+                        "of sub-chunk d+1 overlaps the reduce of d "
+                        "(1 = the unpipelined three-stage form)")
     p.add_argument("--no-check", action="store_true",
                    help="skip verifying the hand-rolled algorithms "
                         "against RCCL before timing")
@@ -95,7 +99,7 @@ def main():
         dist.all_reduce(ref, op=dist.ReduceOp.SUM)
         for algo in ("fullmesh", "ring"):
             t = base.clone()
-            xgmi_all_reduce(t, algo=algo)
+            xgmi_all_reduce(t, algo=algo, depth=args.pipeline_depth)
             torch.cuda.synchronize()
             ok = torch.allclose(t.float(), ref.float(),
                                 atol=1e-2 if dtype == torch.bfloat16
@@ -110,8 +114,11 @@ def main():
         if world == 1 and algo != "rccl":
             continue  # p2p algorithms need peers
         t = base.clone()
-        el, bus, alg = run_algo(algo, t, args.iters, args.warmup, world)
+        el, bus, alg = run_algo(algo, t, args.iters, args.warmup, world,
+                                depth=args.pipeline_depth)
         results[algo] = {"ms": el * 1e3, "bus_GBps": bus, "alg_GBps": alg}
+        if algo == "fullmesh":
+            results[algo]["pipeline_depth"] = args.pipeline_depth
 
     if rank == 0:
         print(json.dumps({

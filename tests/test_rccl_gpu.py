@@ -95,6 +95,38 @@ def test_world1_barrier(world1):
     dist.barrier()
 
 
+def test_world1_sendrecv_loopback(world1):
+    """Grouped paired exchange (the deadlock-free ring step) —
+    world-1 loopback proves the native group_start/send/recv/group_end
+    path end to end."""
+    t = torch.randn(4096, device=DEV)
+    r = torch.zeros_like(t)
+    dist.sendrecv(t, 0, r, 0)
+    assert torch.equal(r, t)
+
+
+def test_world1_reduce_columns_streams(world1):
+    """The pipelined fullmesh building block: reduce_columns on the
+    compute stream consuming data ordered by events from a second
+    stream (the schedule of fullmesh_all_reduce at depth>1)."""
+    k = load_native("_kernels")
+    n, P = 1 << 18, 7
+    dst = torch.randn(n, device=DEV)
+    expect = dst.clone()
+    scratch = torch.randn(P, n, device=DEV)
+    expect += scratch.sum(0)
+    comm_s = torch.cuda.Stream()
+    with torch.cuda.stream(comm_s):
+        scratch.mul_(1.0)  # some comm-stream work producing scratch
+        ev = torch.cuda.Event()
+        ev.record(comm_s)
+    torch.cuda.current_stream().wait_event(ev)
+    k.reduce_columns(dst.data_ptr(), scratch.data_ptr(), P, n, n, 1.0,
+                     7, torch.cuda.current_stream().cuda_stream)
+    torch.cuda.synchronize()
+    assert torch.allclose(dst, expect, rtol=1e-5, atol=1e-5)
+
+
 def test_world1_training_step(world1):
     """One full DP training step through the native backend + HIP ops."""
     from dist_tuto_pth_amd import ops
