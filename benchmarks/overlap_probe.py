@@ -5,14 +5,16 @@
 At 256 MB / 8 ranks each owned chunk is 32 MB; the pipeline overlaps
 the grouped xGMI exchange of sub-chunk d+1 with the
 ``reduce_columns`` fold of sub-chunk d.  On one GPU the exchange is
-stood in for by a same-sized DtoD copy (an xGMI transfer and a local
-HBM copy are both stream-ordered async ops from the scheduling point
-of view): this probe runs the reduce kernel on the compute stream
-while the copy runs on a second stream and reports serial vs
-overlapped wall-clock.  Run it under rocprofv3 --kernel-trace for the
-timeline evidence kept in profiles/:
+stood in for by a *grid-throttled* copy kernel: RCCL moves an xGMI
+peer chunk with a handful of workgroups at the ~153 GB/s link rate, so
+the right single-GPU model is a copy capped well below HBM rate (a
+full-rate DtoD memcpy would itself saturate HBM and leave nothing for
+the reduce to overlap with — measured in round 2 and kept as the
+``--full-rate`` variant).  The probe reports serial vs two-stream
+wall-clock; run it under rocprofv3 --kernel-trace for the timeline
+kept in profiles/:
 
-  rocprofv3 --kernel-trace --stats -d gpurun_out/prof -- \
+  rocprofv3 --kernel-trace --stats -d gpurun_out/prof_overlap -- \
       python benchmarks/overlap_probe.py
 """
 
@@ -35,6 +37,14 @@ def main():
                    help="owned-chunk size (256 MB / 8 ranks default)")
     p.add_argument("--peers", type=int, default=7)
     p.add_argument("--iters", type=int, default=20)
+    p.add_argument("--copy-blocks", type=int, default=24,
+                   help="workgroups for the throttled copy (approximates "
+                        "the per-link xGMI rate; 24 blocks ~ a few hundred "
+                        "GB/s)")
+    p.add_argument("--full-rate", action="store_true",
+                   help="use an unthrottled full-HBM-rate copy instead "
+                        "(shows why that is the wrong model: both ops "
+                        "are HBM-bound, overlap cannot win)")
     args = p.parse_args()
 
     k = load_native("_kernels")
@@ -45,8 +55,8 @@ def main():
 
     dst = torch.randn(n, device=dev)
     scratch = torch.randn(P, n, device=dev)
-    # stand-in for the next sub-chunk's exchange: same bytes on the move
-    cp_src = torch.randn(P * n, device=dev)
+    # stand-in for one peer sub-chunk on the wire
+    cp_src = torch.randn(n, device=dev)
     cp_dst = torch.empty_like(cp_src)
 
     s_compute = torch.cuda.current_stream()
@@ -57,13 +67,20 @@ def main():
                          1.0, 7, s_compute.cuda_stream)
 
     def copy_op(stream):
-        with torch.cuda.stream(stream):
-            cp_dst.copy_(cp_src, non_blocking=True)
+        if args.full_rate:
+            with torch.cuda.stream(stream):
+                cp_dst.copy_(cp_src, non_blocking=True)
+        else:
+            k.copy_throttled(cp_dst.data_ptr(), cp_src.data_ptr(), n,
+                             args.copy_blocks, stream.cuda_stream)
 
-    # warmup
+    # warm up BOTH streams (the second stream's first use pays a
+    # multi-ms lazy HSA-queue creation that poisoned the round-2 first
+    # measurement)
     for _ in range(5):
         reduce_op()
         copy_op(s_compute)
+        copy_op(s_comm)
     torch.cuda.synchronize()
 
     def timed(fn):
@@ -78,13 +95,13 @@ def main():
     t_copy = timed(lambda: copy_op(s_compute))
     t_serial = timed(lambda: (reduce_op(), copy_op(s_compute)))
     t_overlap = timed(lambda: (copy_op(s_comm), reduce_op()))
-    # ensure both streams drained between iterations is handled by the
-    # synchronize bracketing; within an iteration the two ops share no
-    # buffers so no event is needed
 
+    bytes_copy = n * 4
     print(json.dumps({
-        "probe": "reduce_columns vs DtoD copy, two streams",
+        "probe": "reduce_columns vs throttled copy, two streams",
         "chunk_mb": args.chunk_mb, "peers": P, "iters": args.iters,
+        "copy_blocks": (0 if args.full_rate else args.copy_blocks),
+        "copy_GBps": 2 * bytes_copy / (t_copy * 1e-3) / 1e9,
         "reduce_ms": t_reduce, "copy_ms": t_copy,
         "serial_ms": t_serial, "overlapped_ms": t_overlap,
         "overlap_saving_pct":

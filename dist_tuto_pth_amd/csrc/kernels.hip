@@ -782,6 +782,26 @@ __global__ void scale_f32_kernel(float* __restrict__ dst, float s,
     dst[i] *= s;
 }
 
+// grid-throttled float4 copy: a stand-in for a link-bound transfer in
+// single-GPU overlap probes.  RCCL moves an xGMI peer copy with a few
+// workgroups at ~153 GB/s per link — a full-rate DtoD memcpy (5+ TB/s,
+// all of HBM) is the WRONG model for it.  Limiting the grid caps the
+// copy's HBM draw so the concurrent reduce kernel has headroom, which
+// is exactly the real pipeline's situation.
+__global__ void copy_throttled_f32_kernel(float* __restrict__ dst,
+                                          const float* __restrict__ src,
+                                          int64_t n) {
+  const int64_t n4 = n / 4;
+  float4* d4 = reinterpret_cast<float4*>(dst);
+  const float4* s4 = reinterpret_cast<const float4*>(src);
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+       i += (int64_t)gridDim.x * blockDim.x)
+    d4[i] = s4[i];
+  for (int64_t i = n4 * 4 + blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    dst[i] = src[i];
+}
+
 // ===========================================================================
 // Fused whole-Net forward / backward (the launch-bound-regime lever).
 //
@@ -2407,6 +2427,12 @@ void scale_f32(uintptr_t dst, double s, int64_t n, uintptr_t stream) {
                      0, S(stream), (float*)dst, (float)s, n);
 }
 
+void copy_throttled(uintptr_t dst, uintptr_t src, int64_t n, int nblocks,
+                    uintptr_t stream) {
+  hipLaunchKernelGGL(copy_throttled_f32_kernel, dim3(nblocks), dim3(BLK),
+                     0, S(stream), (float*)dst, (const float*)src, n);
+}
+
 }  // namespace
 
 PYBIND11_MODULE(_kernels, m) {
@@ -2441,6 +2467,7 @@ PYBIND11_MODULE(_kernels, m) {
   m.def("flag_barrier_probe", &flag_barrier_probe);
   m.def("net_fused_bwd_sgd", &net_fused_bwd_sgd);
   m.def("add_inplace", &add_inplace);
+  m.def("copy_throttled", &copy_throttled);
   m.def("reduce_columns", &reduce_columns);
   m.def("scale_f32", &scale_f32);
 }
