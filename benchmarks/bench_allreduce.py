@@ -26,6 +26,48 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 from dist_tuto_pth_amd import dist  # noqa: E402
 from dist_tuto_pth_amd.algorithms.xgmi import xgmi_all_reduce  # noqa: E402
+from dist_tuto_pth_amd.algorithms.ipc import (  # noqa: E402
+    IpcTransport, fullmesh_all_reduce_ipc)
+from dist_tuto_pth_amd.utils.native import load_native  # noqa: E402
+
+_TP = [None]
+
+
+def _ipc_transport(world, rank, dev_idx, numel, esz):
+    if _TP[0] is None:
+        store = dist.group.WORLD._impl._store
+        chunk_cap = ((numel + world - 1) // world + 64) * esz
+        _TP[0] = IpcTransport(store, rank, world, chunk_cap,
+                              device=dev_idx, tag="benchp2p")
+    return _TP[0]
+
+
+def run_p2p(t, iters, warmup, world, rank, dev_idx):
+    """--algos p2p: the self-owned transport — IPC mesh + one-sided
+    hipMemcpyAsync pushes over xGMI, no RCCL on the wire."""
+    k = load_native("_kernels")
+    tp = _ipc_transport(world, rank, dev_idx, t.numel(), t.element_size())
+
+    def op():
+        fullmesh_all_reduce_ipc(t, tp, k, rank, world)
+
+    for _ in range(warmup):
+        op()
+    dist.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        op()
+    torch.cuda.synchronize()
+    dist.barrier()
+    el = (time.perf_counter() - t0) / iters
+    e = torch.tensor([el], device=t.device)
+    dist.all_reduce(e, op=dist.ReduceOp.MAX)
+    el = float(e.item())
+    bytes_ = t.numel() * t.element_size()
+    bus_bw = 2 * (world - 1) / world * bytes_ / el / 1e9
+    alg_bw = bytes_ / el / 1e9
+    return el, bus_bw, alg_bw
 
 
 def run_algo(algo, t, iters, warmup, world, depth=4):
@@ -97,9 +139,16 @@ def main():
     if not args.no_check and world > 1:
         ref = base.clone()
         dist.all_reduce(ref, op=dist.ReduceOp.SUM)
-        for algo in ("fullmesh", "ring"):
+        algos = args.algos.split(",")
+        for algo in [a for a in ("fullmesh", "ring", "p2p") if a in algos]:
             t = base.clone()
-            xgmi_all_reduce(t, algo=algo, depth=args.pipeline_depth)
+            if algo == "p2p":
+                fullmesh_all_reduce_ipc(
+                    t, _ipc_transport(world, rank, dev_idx, t.numel(),
+                                      t.element_size()),
+                    load_native("_kernels"), rank, world)
+            else:
+                xgmi_all_reduce(t, algo=algo, depth=args.pipeline_depth)
             torch.cuda.synchronize()
             ok = torch.allclose(t.float(), ref.float(),
                                 atol=1e-2 if dtype == torch.bfloat16
@@ -114,8 +163,12 @@ def main():
         if world == 1 and algo != "rccl":
             continue  # p2p algorithms need peers
         t = base.clone()
-        el, bus, alg = run_algo(algo, t, args.iters, args.warmup, world,
-                                depth=args.pipeline_depth)
+        if algo == "p2p":
+            el, bus, alg = run_p2p(t, args.iters, args.warmup, world,
+                                   rank, dev_idx)
+        else:
+            el, bus, alg = run_algo(algo, t, args.iters, args.warmup, world,
+                                    depth=args.pipeline_depth)
         results[algo] = {"ms": el * 1e3, "bus_GBps": bus, "alg_GBps": alg}
         if algo == "fullmesh":
             results[algo]["pipeline_depth"] = args.pipeline_depth

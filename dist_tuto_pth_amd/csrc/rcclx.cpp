@@ -458,6 +458,68 @@ class Comm {
 };
 
 // ---------------------------------------------------------------------------
+// IPC peer-copy transport (the self-owned wire of the hand-tuned
+// all-reduce — BASELINE's "hipMemcpyPeerAsync across the 7 xGMI links").
+// A DeviceBuffer is a raw hipMalloc allocation (NOT torch caching-
+// allocator memory: hipIpcGetMemHandle must see the real allocation
+// base), exported through hipIpcGetMemHandle; peers exchange the
+// 64-byte handles through the TCP store and open them with
+// hipIpcOpenMemHandle, after which a plain stream-ordered
+// hipMemcpyAsync into the mapped peer pointer IS the xGMI push.
+// Requires dmabuf IPC (HSA_ENABLE_IPC_MODE_LEGACY=0, exported by the
+// environment).
+// ---------------------------------------------------------------------------
+class DeviceBuffer {
+ public:
+  DeviceBuffer(size_t nbytes, int device) : nbytes_(nbytes) {
+    HIP_CHECK(hipSetDevice(device));
+    HIP_CHECK(hipMalloc(&ptr_, nbytes));
+  }
+  ~DeviceBuffer() {
+    if (ptr_) hipFree(ptr_);
+  }
+  uintptr_t ptr() const { return reinterpret_cast<uintptr_t>(ptr_); }
+  size_t nbytes() const { return nbytes_; }
+  py::bytes ipc_handle() const {
+    hipIpcMemHandle_t h;
+    HIP_CHECK(hipIpcGetMemHandle(&h, ptr_));
+    return py::bytes(reinterpret_cast<const char*>(&h), sizeof(h));
+  }
+
+ private:
+  void* ptr_ = nullptr;
+  size_t nbytes_;
+};
+
+static uintptr_t ipc_open(const py::bytes& handle_bytes) {
+  std::string s = handle_bytes;
+  if (s.size() != sizeof(hipIpcMemHandle_t))
+    throw std::runtime_error("bad hipIpcMemHandle size");
+  hipIpcMemHandle_t h;
+  memcpy(&h, s.data(), sizeof(h));
+  void* p = nullptr;
+  py::gil_scoped_release nogil;
+  HIP_CHECK(hipIpcOpenMemHandle(&p, h, hipIpcMemLazyEnablePeerAccess));
+  return reinterpret_cast<uintptr_t>(p);
+}
+
+static void ipc_close(uintptr_t ptr) {
+  py::gil_scoped_release nogil;
+  HIP_CHECK(hipIpcCloseMemHandle(reinterpret_cast<void*>(ptr)));
+}
+
+static void memcpy_async(uintptr_t dst, uintptr_t src, size_t nbytes,
+                         uintptr_t stream) {
+  // unified addressing: works for local DtoD and for IPC-mapped peer
+  // destinations (the xGMI push)
+  py::gil_scoped_release nogil;
+  HIP_CHECK(hipMemcpyAsync(reinterpret_cast<void*>(dst),
+                           reinterpret_cast<void*>(src), nbytes,
+                           hipMemcpyDeviceToDevice,
+                           reinterpret_cast<hipStream_t>(stream)));
+}
+
+// ---------------------------------------------------------------------------
 // small HIP utilities for the Python side
 // ---------------------------------------------------------------------------
 static py::bytes get_unique_id() {
@@ -523,6 +585,16 @@ PYBIND11_MODULE(_rcclx, m) {
       .def("rank", &Comm::rank)
       .def("nranks", &Comm::nranks)
       .def("device", &Comm::device);
+
+  py::class_<DeviceBuffer>(m, "DeviceBuffer")
+      .def(py::init<size_t, int>(), py::arg("nbytes"), py::arg("device"))
+      .def("ptr", &DeviceBuffer::ptr)
+      .def("nbytes", &DeviceBuffer::nbytes)
+      .def("ipc_handle", &DeviceBuffer::ipc_handle);
+
+  m.def("ipc_open", &ipc_open);
+  m.def("ipc_close", &ipc_close);
+  m.def("memcpy_async", &memcpy_async);
 
   m.def("get_unique_id", &get_unique_id);
   m.def("record_event", &record_event);
