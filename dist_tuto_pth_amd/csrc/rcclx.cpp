@@ -26,6 +26,9 @@
 #include <sys/socket.h>
 #include <unistd.h>
 
+#include <sys/select.h>
+
+#include <algorithm>
 #include <atomic>
 #include <chrono>
 #include <condition_variable>
@@ -458,6 +461,182 @@ class Comm {
 };
 
 // ---------------------------------------------------------------------------
+// TcpMesh — the full-mesh CPU p2p layer of the native "tcp" backend.
+// This is the part of THD the tutorial documents in prose
+// (tuto.md:409-419): every pair of ranks holds one TCP connection,
+// established master/worker-style (rank i accepts from all j > i and
+// connects to all j < i; addresses travel through the TCP store).
+// Messages are length-framed; targeted recv reads the pair socket
+// (per-peer FIFO ordering), any-source recv polls all sockets.  Reads
+// and writes move raw bytes to/from caller-provided pointers (zero
+// copy from contiguous CPU tensors).
+// ---------------------------------------------------------------------------
+class TcpMesh {
+ public:
+  TcpMesh(TcpStore* store, const std::string& tag, int rank, int world,
+          int timeout_ms)
+      : rank_(rank), world_(world) {
+    fds_.assign(world, -1);
+    send_mu_ = std::vector<std::mutex>(world);
+    recv_mu_ = std::vector<std::mutex>(world);
+    if (world == 1) return;
+    // bind an ephemeral listening socket and publish its port
+    int lfd = ::socket(AF_INET, SOCK_STREAM, 0);
+    if (lfd < 0) throw std::runtime_error("mesh: socket() failed");
+    int one = 1;
+    setsockopt(lfd, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+    sockaddr_in addr{};
+    addr.sin_family = AF_INET;
+    addr.sin_port = 0;
+    addr.sin_addr.s_addr = INADDR_ANY;
+    if (::bind(lfd, reinterpret_cast<sockaddr*>(&addr), sizeof(addr)))
+      throw std::runtime_error("mesh: bind failed");
+    socklen_t alen = sizeof(addr);
+    getsockname(lfd, reinterpret_cast<sockaddr*>(&addr), &alen);
+    int my_port = ntohs(addr.sin_port);
+    if (::listen(lfd, world))
+      throw std::runtime_error("mesh: listen failed");
+    {
+      py::gil_scoped_acquire gil;  // store API touches py::bytes
+      store->set(tag + ":addr:" + std::to_string(rank),
+                 py::bytes(std::to_string(my_port)));
+    }
+    // connect to lower ranks (each connection opens with our rank id)
+    for (int j = 0; j < rank; ++j) {
+      std::string port_s;
+      {
+        py::gil_scoped_acquire gil;
+        port_s = std::string(store->get(tag + ":addr:" + std::to_string(j)));
+      }
+      int port = std::stoi(port_s);
+      auto deadline = std::chrono::steady_clock::now() +
+                      std::chrono::milliseconds(timeout_ms);
+      int fd;
+      for (;;) {
+        fd = ::socket(AF_INET, SOCK_STREAM, 0);
+        sockaddr_in a{};
+        a.sin_family = AF_INET;
+        a.sin_port = htons(static_cast<uint16_t>(port));
+        a.sin_addr.s_addr = htonl(INADDR_LOOPBACK);
+        if (::connect(fd, reinterpret_cast<sockaddr*>(&a), sizeof(a)) == 0)
+          break;
+        ::close(fd);
+        if (std::chrono::steady_clock::now() > deadline)
+          throw std::runtime_error("mesh: connect to rank " +
+                                   std::to_string(j) + " timed out");
+        std::this_thread::sleep_for(std::chrono::milliseconds(20));
+      }
+      setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+      uint32_t me = static_cast<uint32_t>(rank);
+      send_all(fd, &me, 4);
+      fds_[j] = fd;
+    }
+    // accept from higher ranks
+    for (int n = 0; n < world - 1 - rank; ++n) {
+      int fd = ::accept(lfd, nullptr, nullptr);
+      if (fd < 0) throw std::runtime_error("mesh: accept failed");
+      setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+      uint32_t peer;
+      if (!recv_all(fd, &peer, 4))
+        throw std::runtime_error("mesh: peer id read failed");
+      if (peer >= static_cast<uint32_t>(world) || fds_[peer] != -1)
+        throw std::runtime_error("mesh: bad peer id");
+      fds_[peer] = fd;
+    }
+    ::close(lfd);
+  }
+
+  ~TcpMesh() {
+    for (int fd : fds_)
+      if (fd >= 0) ::close(fd);
+  }
+
+  void send(int peer, uintptr_t ptr, uint64_t nbytes) {
+    py::gil_scoped_release nogil;
+    std::lock_guard<std::mutex> lk(send_mu_[peer]);
+    int fd = fds_.at(peer);
+    if (fd < 0) throw std::runtime_error("mesh: no link to peer");
+    send_all(fd, &nbytes, 8);
+    if (nbytes) send_all(fd, reinterpret_cast<void*>(ptr), nbytes);
+  }
+
+  void recv(int peer, uintptr_t ptr, uint64_t nbytes) {
+    py::gil_scoped_release nogil;
+    std::lock_guard<std::mutex> lk(recv_mu_[peer]);
+    int fd = fds_.at(peer);
+    if (fd < 0) throw std::runtime_error("mesh: no link to peer");
+    read_frame(fd, ptr, nbytes, peer);
+  }
+
+  // any-source receive (tuto.md:90 semantics): poll every pair socket,
+  // take the first message that arrives.  Returns the source rank.
+  int recv_any(uintptr_t ptr, uint64_t nbytes, int timeout_ms) {
+    py::gil_scoped_release nogil;
+    auto deadline = std::chrono::steady_clock::now() +
+                    std::chrono::milliseconds(timeout_ms);
+    for (;;) {
+      fd_set rset;
+      FD_ZERO(&rset);
+      int maxfd = -1;
+      for (int p = 0; p < world_; ++p) {
+        if (fds_[p] >= 0) {
+          FD_SET(fds_[p], &rset);
+          maxfd = std::max(maxfd, fds_[p]);
+        }
+      }
+      timeval tv{0, 200 * 1000};  // 200 ms poll slices
+      int n = ::select(maxfd + 1, &rset, nullptr, nullptr, &tv);
+      if (n > 0) {
+        for (int p = 0; p < world_; ++p) {
+          if (fds_[p] >= 0 && FD_ISSET(fds_[p], &rset)) {
+            std::unique_lock<std::mutex> lk(recv_mu_[p],
+                                            std::try_to_lock);
+            if (!lk.owns_lock()) continue;  // someone else is reading p
+            // a peer that exited leaves an EOF-readable socket: peek
+            // the header so a drained link is skipped, not mistaken
+            // for a message
+            uint64_t hdr;
+            ssize_t k = ::recv(fds_[p], &hdr, 8, MSG_PEEK);
+            if (k == 0) {
+              ::close(fds_[p]);
+              fds_[p] = -1;
+              continue;
+            }
+            if (k < 8) continue;  // header still arriving
+            read_frame(fds_[p], ptr, nbytes, p);
+            return p;
+          }
+        }
+      }
+      if (std::chrono::steady_clock::now() > deadline)
+        throw std::runtime_error("mesh: recv(any) timed out");
+    }
+  }
+
+  int rank() const { return rank_; }
+  int world() const { return world_; }
+
+ private:
+  void read_frame(int fd, uintptr_t ptr, uint64_t nbytes, int peer) {
+    uint64_t len;
+    if (!recv_all(fd, &len, 8))
+      throw std::runtime_error("mesh: recv header failed from rank " +
+                               std::to_string(peer));
+    if (len != nbytes)
+      throw std::runtime_error(
+          "mesh: size mismatch from rank " + std::to_string(peer) +
+          ": sent " + std::to_string(len) + "B, receiving " +
+          std::to_string(nbytes) + "B");
+    if (len && !recv_all(fd, reinterpret_cast<void*>(ptr), len))
+      throw std::runtime_error("mesh: recv payload failed");
+  }
+
+  int rank_, world_;
+  std::vector<int> fds_;
+  std::vector<std::mutex> send_mu_, recv_mu_;
+};
+
+// ---------------------------------------------------------------------------
 // IPC peer-copy transport (the self-owned wire of the hand-tuned
 // all-reduce — BASELINE's "hipMemcpyPeerAsync across the 7 xGMI links").
 // A DeviceBuffer is a raw hipMalloc allocation (NOT torch caching-
@@ -476,7 +655,7 @@ class DeviceBuffer {
     HIP_CHECK(hipMalloc(&ptr_, nbytes));
   }
   ~DeviceBuffer() {
-    if (ptr_) hipFree(ptr_);
+    if (ptr_) (void)hipFree(ptr_);
   }
   uintptr_t ptr() const { return reinterpret_cast<uintptr_t>(ptr_); }
   size_t nbytes() const { return nbytes_; }
@@ -585,6 +764,18 @@ PYBIND11_MODULE(_rcclx, m) {
       .def("rank", &Comm::rank)
       .def("nranks", &Comm::nranks)
       .def("device", &Comm::device);
+
+  py::class_<TcpMesh>(m, "TcpMesh")
+      .def(py::init<TcpStore*, const std::string&, int, int, int>(),
+           py::arg("store"), py::arg("tag"), py::arg("rank"),
+           py::arg("world"), py::arg("timeout_ms") = 120000,
+           py::keep_alive<1, 2>())
+      .def("send", &TcpMesh::send)
+      .def("recv", &TcpMesh::recv)
+      .def("recv_any", &TcpMesh::recv_any, py::arg("ptr"),
+           py::arg("nbytes"), py::arg("timeout_ms") = 120000)
+      .def("rank", &TcpMesh::rank)
+      .def("world", &TcpMesh::world);
 
   py::class_<DeviceBuffer>(m, "DeviceBuffer")
       .def(py::init<size_t, int>(), py::arg("nbytes"), py::arg("device"))

@@ -223,16 +223,17 @@ def init_process_group(
     if _state.backend is not None:
         raise RuntimeError("process group already initialized")
     backend = backend.lower()
-    if backend == "tcp":
-        # the reference's 'tcp' THD backend (ptp.py:30) — CPU-only p2p +
-        # collectives; gloo is its modern CPU equivalent here.
-        backend = "gloo"
     if backend == "nccl":
         backend = "rccl"
 
     addr, port, world_size, rank = _parse_init(init_method, world_size, rank)
 
-    if backend == "gloo":
+    if backend == "tcp":
+        # the reference's 'tcp' THD backend (ptp.py:30), owned natively:
+        # per-pair sockets + p2p-composed collectives (tcp_backend.py).
+        impl = _TcpBackend(addr, port, world_size, rank)
+    elif backend == "gloo":
+        # torch.distributed delegation, kept as a CPU cross-check only
         impl = _GlooBackend(addr, port, world_size, rank)
     elif backend == "rccl":
         impl = _RcclBackend(addr, port, world_size, rank, device_id)
@@ -576,6 +577,8 @@ class _GlooBackend:
 
 
 # --------------------------------------------------------------------------
-# RCCL backend (the MI355X path) — implemented in csrc/rcclx.cpp
+# native backends (csrc/rcclx.cpp): RCCL-over-xGMI (the MI355X path) and
+# the self-owned CPU "tcp" mesh
 # --------------------------------------------------------------------------
 from .rccl_backend import _RcclBackend  # noqa: E402  (needs Work/ReduceOp above)
+from .tcp_backend import _TcpBackend  # noqa: E402

@@ -65,7 +65,8 @@ def run(rank, size):
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--world", type=int, default=4)
-    p.add_argument("--backend", default="gloo", choices=["gloo", "rccl"])
+    p.add_argument("--backend", default="gloo",
+                   choices=["tcp", "gloo", "rccl"])
     p.add_argument("--algo", default="chunked",
                    choices=["ring", "chunked", "fullmesh"])
     p.add_argument("--numel", type=int, default=1000)

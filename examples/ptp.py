@@ -8,7 +8,7 @@ instead of the 0.x positional-group form (ptp.py:26), plus the legacy
 root-split pair ``gather_recv``/``gather_send`` (ptp.py:17-19) shown in
 ``run_legacy_pair``.
 
-Run:  python examples/ptp.py [--world 2] [--backend gloo|rccl]
+Run:  python examples/ptp.py [--world 2] [--backend tcp|gloo|rccl]
 """
 
 import argparse
@@ -50,10 +50,11 @@ def run_legacy_pair(rank, size):
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--world", type=int, default=2)
-    p.add_argument("--backend", default="gloo", choices=["gloo", "rccl"])
+    p.add_argument("--backend", default="tcp",
+                   choices=["tcp", "gloo", "rccl"])
     args = p.parse_args()
     launch(run, args.world, backend=args.backend)
-    if args.backend == "gloo":
+    if args.backend != "rccl":
         launch(run_legacy_pair, args.world, backend=args.backend)
 
 

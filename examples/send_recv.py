@@ -4,7 +4,7 @@ single fp32 scalar between two ranks, then the non-blocking
 isend/irecv + wait() form.  This is BASELINE.md config 1 (CPU plumbing,
 no GPU required).
 
-Run:  python examples/send_recv.py [--backend gloo|rccl]
+Run:  python examples/send_recv.py [--backend tcp|gloo|rccl]
 """
 
 import argparse
@@ -21,7 +21,7 @@ from dist_tuto_pth_amd.dist.launcher import launch
 
 
 def run_blocking(rank, size):
-    device = "cpu" if dist.get_backend() == "gloo" \
+    device = "cpu" if dist.get_backend() in ("gloo", "tcp") \
         else f"cuda:{torch.cuda.current_device()}"
     tensor = torch.zeros(1, device=device)
     if rank == 0:
@@ -72,10 +72,11 @@ def run_latency(rank, size, iters=1000):
 
 def main():
     p = argparse.ArgumentParser()
-    p.add_argument("--backend", default="gloo", choices=["gloo", "rccl"])
+    p.add_argument("--backend", default="tcp",
+                   choices=["tcp", "gloo", "rccl"])
     args = p.parse_args()
     launch(run_blocking, 2, backend=args.backend)
-    if args.backend == "gloo":
+    if args.backend != "rccl":
         launch(run_nonblocking, 2, backend=args.backend)
         launch(run_latency, 2, backend=args.backend)
 

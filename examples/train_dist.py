@@ -40,7 +40,8 @@ def run(rank, size):
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--world", type=int, default=2)
-    p.add_argument("--backend", default="gloo", choices=["gloo", "rccl"])
+    p.add_argument("--backend", default="gloo",
+                   choices=["tcp", "gloo", "rccl"])
     p.add_argument("--epochs", type=int, default=2)
     p.add_argument("--mode", default="average_gradients",
                    choices=["average_gradients", "ddp"])

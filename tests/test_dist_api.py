@@ -1,8 +1,12 @@
-"""API-surface tests over the gloo CPU path, fork-local-ranks over
+"""API-surface tests over the CPU backends, fork-local-ranks over
 loopback (the reference's own harness shape): every L2
-primitive the tutorial catalogues (tuto.md:77-202)."""
+primitive the tutorial catalogues (tuto.md:77-202), run on BOTH the
+native "tcp" backend (this repo's own sockets, tcp_backend.py) and the
+gloo cross-check delegation."""
 
 import os
+
+import pytest
 import torch
 
 from dist_tuto_pth_amd import dist
@@ -127,7 +131,7 @@ def _fn_rank_world(rank, size):
     assert dist.get_rank() == rank
     assert dist.get_world_size() == size
     assert dist.is_initialized()
-    assert dist.get_backend() == "gloo"
+    assert dist.get_backend() in ("gloo", "tcp")
     dist.barrier()
 
 
@@ -142,48 +146,59 @@ def _fn_ptp_demo(rank, size):
 
 # ---- drivers ------------------------------------------------------------
 
-def test_ptp_blocking():
-    launch(_fn_ptp_blocking, 2)
+@pytest.mark.parametrize("backend", ["gloo", "tcp"])
+def test_ptp_blocking(backend):
+    launch(_fn_ptp_blocking, 2, backend=backend)
 
 
-def test_ptp_nonblocking():
-    launch(_fn_ptp_nonblocking, 2)
+@pytest.mark.parametrize("backend", ["gloo", "tcp"])
+def test_ptp_nonblocking(backend):
+    launch(_fn_ptp_nonblocking, 2, backend=backend)
 
 
-def test_all_reduce_ops():
-    launch(_fn_all_reduce, 2)
+@pytest.mark.parametrize("backend", ["gloo", "tcp"])
+def test_all_reduce_ops(backend):
+    launch(_fn_all_reduce, 2, backend=backend)
 
 
-def test_broadcast_reduce():
-    launch(_fn_broadcast_reduce, 2)
+@pytest.mark.parametrize("backend", ["gloo", "tcp"])
+def test_broadcast_reduce(backend):
+    launch(_fn_broadcast_reduce, 2, backend=backend)
 
 
-def test_gather_scatter():
-    launch(_fn_gather_scatter, 2)
+@pytest.mark.parametrize("backend", ["gloo", "tcp"])
+def test_gather_scatter(backend):
+    launch(_fn_gather_scatter, 2, backend=backend)
 
 
-def test_all_gather():
-    launch(_fn_all_gather, 2)
+@pytest.mark.parametrize("backend", ["gloo", "tcp"])
+def test_all_gather(backend):
+    launch(_fn_all_gather, 2, backend=backend)
 
 
-def test_reduce_scatter_alltoall():
-    launch(_fn_reduce_scatter_alltoall, 2)
+@pytest.mark.parametrize("backend", ["gloo", "tcp"])
+def test_reduce_scatter_alltoall(backend):
+    launch(_fn_reduce_scatter_alltoall, 2, backend=backend)
 
 
-def test_gather_send_recv_pair():
-    launch(_fn_gather_pair, 2)
+@pytest.mark.parametrize("backend", ["gloo", "tcp"])
+def test_gather_send_recv_pair(backend):
+    launch(_fn_gather_pair, 2, backend=backend)
 
 
-def test_new_group():
-    launch(_fn_new_group, 2)
+@pytest.mark.parametrize("backend", ["gloo", "tcp"])
+def test_new_group(backend):
+    launch(_fn_new_group, 2, backend=backend)
 
 
-def test_rank_world_backend():
-    launch(_fn_rank_world, 2)
+@pytest.mark.parametrize("backend", ["gloo", "tcp"])
+def test_rank_world_backend(backend):
+    launch(_fn_rank_world, 2, backend=backend)
 
 
-def test_ptp_demo_world2():
-    launch(_fn_ptp_demo, 2)
+@pytest.mark.parametrize("backend", ["gloo", "tcp"])
+def test_ptp_demo_world2(backend):
+    launch(_fn_ptp_demo, 2, backend=backend)
 
 
 # ---------------------------------------------------------------------------
