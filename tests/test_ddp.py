@@ -62,8 +62,58 @@ def _fn_avg_matches_manual(rank, size):
         assert torch.allclose(lg, p.grad, atol=1e-6)
 
 
+def _fn_ddp_zero_copy(rank, size):
+    """Zero-copy contract: every grad is a VIEW of its bucket flat, the
+    wiring survives multiple steps and set_to_none, and repeated steps
+    with ddp.zero_grad() give the same averaged grads as fresh ones."""
+    torch.manual_seed(7)
+    model = Net()
+    model.eval()
+    ddp = DistributedDataParallel(model, bucket_cap_mb=0.01)
+
+    def in_flat(p):
+        for b in ddp.buckets:
+            s = b.flat.data_ptr()
+            e = s + b.flat.numel() * b.flat.element_size()
+            if s <= p.grad.data_ptr() < e:
+                return True
+        return False
+
+    g = torch.Generator().manual_seed(900 + rank)
+    x = torch.randn(4, 1, 28, 28, generator=g)
+    tgt = torch.randint(0, 10, (4,), generator=g)
+
+    out = ddp(x)
+    torch.nn.functional.nll_loss(out, tgt).backward()
+    ddp.finish_gradients()
+    assert all(in_flat(p) for p in model.parameters())
+    first = [p.grad.clone() for p in model.parameters()]
+
+    # second step after zero_grad must reproduce the same grads
+    ddp.zero_grad()
+    out = ddp(x)
+    torch.nn.functional.nll_loss(out, tgt).backward()
+    ddp.finish_gradients()
+    for p, f in zip(model.parameters(), first):
+        assert torch.allclose(p.grad, f, atol=1e-6)
+
+    # an optimizer that detaches grads is healed by the re-attach guard
+    for p in model.parameters():
+        p.grad = None
+    out = ddp(x)  # forward re-attaches
+    torch.nn.functional.nll_loss(out, tgt).backward()
+    ddp.finish_gradients()
+    assert all(in_flat(p) for p in model.parameters())
+    for p, f in zip(model.parameters(), first):
+        assert torch.allclose(p.grad, f, atol=1e-6)
+
+
 def test_ddp_matches_average_gradients():
     launch(_fn_ddp_grads, 2, timeout=300)
+
+
+def test_ddp_zero_copy_views():
+    launch(_fn_ddp_zero_copy, 2, timeout=300)
 
 
 def test_average_gradients_math():
