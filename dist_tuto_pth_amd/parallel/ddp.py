@@ -65,10 +65,15 @@ class _Bucket:
         """Point every parameter's ``.grad`` at its slice of the flat
         buffer (zero-copy: backward accumulates into the bucket)."""
         for p, off, n in zip(self.params, self.offsets, self.numels):
-            if p.grad is None or p.grad.data_ptr() !=                     self.flat[off:off + n].data_ptr():
-                g = self.flat[off:off + n].view_as(p)
+            sl = self.flat[off:off + n]
+            if p.grad is None or p.grad.data_ptr() != sl.data_ptr():
+                g = sl.view_as(p)
                 if p.grad is not None:
                     g.copy_(p.grad.reshape(-1).view_as(p))
+                else:
+                    # grad=None means "accumulate fresh": the slice may
+                    # hold last step's values
+                    g.zero_()
                 p.grad = g
 
 
