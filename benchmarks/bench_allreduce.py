@@ -77,15 +77,19 @@ def run_algo(algo, t, iters, warmup, world, depth=4):
         else:
             xgmi_all_reduce(t, algo=algo, depth=depth)
 
+    def sync():
+        if t.is_cuda:
+            torch.cuda.synchronize()
+
     for _ in range(warmup):
         op()
     if world > 1:
         dist.barrier()
-    torch.cuda.synchronize()
+    sync()
     t0 = time.perf_counter()
     for _ in range(iters):
         op()
-    torch.cuda.synchronize()
+    sync()
     if world > 1:
         dist.barrier()
     el = (time.perf_counter() - t0) / iters
@@ -124,17 +128,29 @@ def main():
         return
     rank = int(os.environ.get("RANK", 0))
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
-    dev_idx = local_rank % torch.cuda.device_count()
-    if world > 1:
-        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        os.environ.setdefault("MASTER_PORT", "29500")
+    # DTP_BENCH_CPU=1: identical launch/arg/rendezvous path on
+    # CPU/gloo so the torchrun contract is testable without a GPU
+    # (tests/test_bench_contract.py); only the built-in all-reduce
+    # runs there.
+    cpu_mode = os.environ.get("DTP_BENCH_CPU") == "1"
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29500")
+    if cpu_mode:
+        dist.init_process_group("gloo", world_size=world, rank=rank)
+        dev_idx = 0
+        args.algos = "rccl"
+        args.no_check = True
+    else:
+        dev_idx = local_rank % torch.cuda.device_count()
         dist.init_process_group("rccl", world_size=world, rank=rank,
                                 device_id=dev_idx)
-    torch.cuda.set_device(dev_idx)
+        torch.cuda.set_device(dev_idx)
     dtype = torch.float32 if args.dtype == "fp32" else torch.bfloat16
     numel = int(args.size_mb * 1e6) // (4 if dtype == torch.float32 else 2)
     g = torch.Generator().manual_seed(42 + rank)
-    base = torch.randn(numel, generator=g).to(dtype).cuda()
+    base = torch.randn(numel, generator=g).to(dtype)
+    if not cpu_mode:
+        base = base.cuda()
 
     if not args.no_check and world > 1:
         ref = base.clone()
