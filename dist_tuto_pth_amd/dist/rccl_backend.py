@@ -106,7 +106,10 @@ class _RcclBackend:
         self._comm = rx.Comm(world_size, rank, uid, dev)
         self._world = world_size
         self._rank = rank
+        self._ptag = "w"   # p2p-notification namespace (world group)
         self._barrier_buf = torch.zeros(1, device="cuda")
+
+    _split_seq = 0
 
     # ------------------------------------------------------------------
     def split(self, ranks):
@@ -114,11 +117,13 @@ class _RcclBackend:
         color = 0 if member else -1     # -1 => NCCL_SPLIT_NOCOLOR
         key = ranks.index(self._rank) if member else 0
         sub = self._comm.split(color, key)
+        _RcclBackend._split_seq += 1
         if not member:
             return _InactiveBackend()
         b = _RcclBackend(None, None, None, None, _comm=sub,
                          _store=self._store, _ws=len(ranks),
                          _rk=ranks.index(self._rank))
+        b._ptag = f"g{_RcclBackend._split_seq}"
         b._barrier_buf = torch.zeros(1, device="cuda")
         return b
 
@@ -128,11 +133,29 @@ class _RcclBackend:
             self._comm = None
 
     # ------------------------------------------------------------------
-    # p2p
+    # p2p.  RCCL matching is by (peer, order), so a source-less receive
+    # (tuto.md:90) cannot be posted directly: every dist-level send
+    # bumps a per-pair sequence counter in the TCP store (a commutative
+    # ADD, so creation order does not matter) and an any-source
+    # receiver polls the peers' next-sequence keys, learns which rank
+    # sent first, then posts the targeted ncclRecv.  Explicit receives
+    # advance the same counters, so the two forms can interleave.  The
+    # store round-trip (~tens of us) rides on the p2p demo path only —
+    # collectives and the hand-tuned algorithms never touch it.
     # ------------------------------------------------------------------
+    def _seq_key(self, dst, src, seq):
+        return f"p2p:n:{self._ptag}:{dst}:{src}:{seq}"
+
     def send(self, t, dst, blocking):
         from . import Work
         _check(t)
+        if self._store is not None:
+            seqs = getattr(self, "_send_seq", None)
+            if seqs is None:
+                seqs = self._send_seq = {}
+            s = seqs.get(dst, 0)
+            seqs[dst] = s + 1
+            self._store.add(self._seq_key(dst, self._rank, s), 1)
         self._comm.send(t.data_ptr(), t.numel(), _DTYPE[t.dtype], dst,
                         _stream())
         if blocking:
@@ -141,12 +164,38 @@ class _RcclBackend:
         ev = self._rx.record_event(_stream())
         return Work(lambda: self._rx.event_wait(ev))
 
+    def _resolve_any_source(self):
+        import time as _time
+        if self._store is None:
+            raise RuntimeError(
+                "any-source recv needs the store (sub-groups inherit it)")
+        seqs = getattr(self, "_recv_seq", None)
+        if seqs is None:
+            seqs = self._recv_seq = {}
+        deadline = _time.time() + 300.0
+        while True:
+            for s in range(self._world):
+                if s == self._rank:
+                    continue
+                nxt = seqs.get(s, 0)
+                if self._store.add(self._seq_key(self._rank, s, nxt),
+                                   0) >= 1:
+                    seqs[s] = nxt + 1
+                    return s
+            if _time.time() > deadline:
+                raise TimeoutError("recv(src=None) timed out")
+            _time.sleep(0.0002)
+
     def recv(self, t, src, blocking):
         from . import Work
         _check(t)
         if src is None:
-            raise ValueError(
-                "rccl backend recv() requires an explicit source rank")
+            src = self._resolve_any_source()
+        else:
+            seqs = getattr(self, "_recv_seq", None)
+            if seqs is None:
+                seqs = self._recv_seq = {}
+            seqs[src] = seqs.get(src, 0) + 1
         self._comm.recv(t.data_ptr(), t.numel(), _DTYPE[t.dtype], src,
                         _stream())
         if blocking:
