@@ -47,6 +47,8 @@ def parse_args():
     p.add_argument("--no-fused", action="store_true",
                    help="use the modular per-op kernel pipeline instead "
                         "of the fused whole-Net kernels")
+    p.add_argument("--fwdbwd", action="store_true",
+                   help="combined fwd+bwd kernel (3-dispatch step)")
     p.add_argument("--megakernel", action="store_true",
                    help="run the whole step as ONE cooperative kernel "
                         "launch (measured slower than the 6-dispatch "
@@ -108,6 +110,7 @@ def main():
     if use_fused:
         from dist_tuto_pth_amd.ops.fused import (attach_flat_grads,
                                                  net_fused_step,
+                                                 net_fused_step_fb,
                                                  net_fused_step_opt,
                                                  net_fused_train_step,
                                                  net_step_available)
@@ -123,6 +126,13 @@ def main():
                 if world == 1:
                     return net_fused_train_step(model, x, tgt, opt)
                 loss = net_fused_train_step(model, x, tgt, do_sgd=False)
+                dist.all_reduce(flat_grads, op=dist.ReduceOp.AVG)
+                opt.step()
+                return loss
+            if args.fwdbwd:
+                if world == 1:
+                    return net_fused_step_fb(model, x, tgt, opt)
+                loss = net_fused_step_fb(model, x, tgt)
                 dist.all_reduce(flat_grads, op=dist.ReduceOp.AVG)
                 opt.step()
                 return loss
@@ -206,6 +216,7 @@ def main():
                 "parallelism": f"dp{world}",
                 "grad_sync": args.mode,
                 "fused": use_fused,
+                "fwdbwd": bool(args.fwdbwd and use_fused),
                 "megakernel": use_mega,
                 "graph": bool(args.graph),
             },

@@ -1248,6 +1248,72 @@ net_fused_bwd_kernel(
   }
 }
 
+// Combined forward + data-backward in ONE kernel (round-2 experiment,
+// VERDICT r1 #9).  In the training loop the loss gradient is the
+// constant 1, so sample b's data backward depends ONLY on sample b's
+// forward — both run in the same workgroup with a __syncthreads where
+// the inter-kernel dependency used to be.  Saves one ~4.5 us dispatch
+// and the w2 re-staging (5000 floats/WG) the separate bwd kernel pays.
+// Cost: the backward loses its `split` sibling-workgroup parallelism
+// (grid = B, not B*split), so at small B the chip is underfilled.
+// The backward LDS carve aliases the forward's dead buffers: total
+// 5020 (w2s, shared by both phases — same [k*10+c]*25 weight layout)
+// + 6848 floats = 46.5 KB/WG -> 3 WGs/CU.
+__global__ void
+__launch_bounds__(256)
+net_fused_fwdbwd_kernel(
+    const float* __restrict__ x,
+    const float* __restrict__ w1, const float* __restrict__ b1,
+    const float* __restrict__ w2, const float* __restrict__ b2,
+    const float* __restrict__ wf1, const float* __restrict__ bf1,
+    const float* __restrict__ wf2, const float* __restrict__ bf2,
+    const int64_t* __restrict__ tgt,
+    float* __restrict__ p1_ws, uint8_t* __restrict__ idx1_ws,
+    uint8_t* __restrict__ m2_ws, float* __restrict__ p2_ws,
+    uint8_t* __restrict__ idx2_ws, float* __restrict__ h1_ws,
+    uint8_t* __restrict__ m3_ws, float* __restrict__ d3_ws,
+    float* __restrict__ logp_ws,
+    float* __restrict__ glog_ws, float* __restrict__ gh1_ws,
+    float* __restrict__ ga2_ws, float* __restrict__ ga1_ws,
+    float* __restrict__ loss_part,
+    const unsigned long long* __restrict__ seed_p, int B, int training) {
+  __shared__ __attribute__((aligned(16))) float w2s[N_C2K * 250 + N_C2K];
+  __shared__ __attribute__((aligned(16))) float pool[6848];
+  const int tid = threadIdx.x;
+  const uint64_t seed = seed_p[0];
+  // forward carve
+  float* xs = pool;              // 784
+  float* w1s = pool + 784;       // 260
+  float* p1 = pool + 1048;       // 1440  (784+260=1044 -> pad to 1048)
+  float* p2 = pool + 2488;       // 320
+  float* d3 = pool + 2808;       // 50
+  float* logits = pool + 2860;   // 10
+  // backward carve (aliased onto the forward's dead buffers)
+  float* gd2p = pool;            // 5120
+  float* gd2 = pool + 5120;      // 1280
+  float* gp2 = pool + 6400;      // 320
+  float* glg = pool + 6720;      // 10
+  float* gd3 = pool + 6732;      // 50
+  float* gh1 = pool + 6784;      // 50 -> 6834 (buffer 6848)
+  const float sc = 1.f / B;      // dLoss == 1 by construction
+
+  float lsum = 0.f;
+  for (int b = blockIdx.x; b < B; b += gridDim.x) {
+    const float lp_t = net_fwd_sample(
+        b, tid, B, training, seed, x, w1, b1, w2, b2, wf1, bf1, wf2, bf2,
+        tgt, p1_ws, idx1_ws, m2_ws, p2_ws, idx2_ws, h1_ws, m3_ws, d3_ws,
+        logp_ws, xs, w1s, p1, w2s, p2, d3, logits);
+    if (tid == 0) lsum += -lp_t / B;
+    __syncthreads();  // fwd LDS dead + this block's global stashes visible
+    net_bwd_sample(b, 0, 1, tid, B, training, sc, wf1, wf2, tgt,
+                   idx1_ws, m2_ws, idx2_ws, h1_ws, m3_ws, logp_ws,
+                   glog_ws, gh1_ws, ga2_ws, ga1_ws,
+                   w2s, glg, gd3, gh1, gp2, gd2, gd2p);
+    __syncthreads();  // bwd LDS dead before the next sample reuses pool
+  }
+  if (tid == 0) loss_part[blockIdx.x] = lsum;
+}
+
 // Per-chunk partial weight gradients for all four layers in ONE launch
 // (replaces 4 kernels + 8 memsets): grid.x walks tile segments
 // [conv2 | fc1 | conv1 | fc2], grid.y is the batch chunk; partials land
@@ -2118,6 +2184,73 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
                          ? (unsigned long long*)seed_dev : nullptr);
 }
 
+// Combined fwd+bwd (one dispatch) + gw partial + combine: a 3-dispatch
+// training step (vs 4 for fwd / bwd / partial / combine).  When prm_v
+// is non-empty the combine also applies the SGD+momentum update
+// (single-GPU: 3 dispatches total including the optimizer).
+void net_fused_fwdbwd(
+    uintptr_t x, uintptr_t w1, uintptr_t b1, uintptr_t w2, uintptr_t b2,
+    uintptr_t wf1, uintptr_t bf1, uintptr_t wf2, uintptr_t bf2,
+    uintptr_t tgt,
+    uintptr_t p1_ws, uintptr_t idx1_ws, uintptr_t m2_ws, uintptr_t p2_ws,
+    uintptr_t idx2_ws, uintptr_t h1_ws, uintptr_t m3_ws, uintptr_t d3_ws,
+    uintptr_t logp_ws, uintptr_t glog_ws, uintptr_t gh1_ws,
+    uintptr_t ga2_ws, uintptr_t ga1_ws, uintptr_t part_ws,
+    const std::vector<uintptr_t>& grd_v,
+    const std::vector<uintptr_t>& prm_v,
+    const std::vector<uintptr_t>& buf_v,
+    double lr, double mu, int B, bool training,
+    uintptr_t loss_part, uintptr_t loss_out, uintptr_t seed_dev,
+    uintptr_t stream) {
+  if (grd_v.size() != 8)
+    throw std::runtime_error("net_fused_fwdbwd: expected 8 grad ptrs");
+  const int nblk = grid_for(B, 1);
+  hipLaunchKernelGGL(net_fused_fwdbwd_kernel, dim3(nblk), dim3(256), 0,
+                     S(stream), (const float*)x, (const float*)w1,
+                     (const float*)b1, (const float*)w2, (const float*)b2,
+                     (const float*)wf1, (const float*)bf1,
+                     (const float*)wf2, (const float*)bf2,
+                     (const int64_t*)tgt, (float*)p1_ws,
+                     (uint8_t*)idx1_ws, (uint8_t*)m2_ws, (float*)p2_ws,
+                     (uint8_t*)idx2_ws, (float*)h1_ws, (uint8_t*)m3_ws,
+                     (float*)d3_ws, (float*)logp_ws, (float*)glog_ws,
+                     (float*)gh1_ws, (float*)ga2_ws, (float*)ga1_ws,
+                     (float*)loss_part,
+                     (const unsigned long long*)seed_dev, B,
+                     training ? 1 : 0);
+  const int bchunk = (B + 31) / 32;
+  const int nch = (B + bchunk - 1) / bchunk;
+  hipLaunchKernelGGL(net_gw_partial_kernel, dim3(GW_TILES, nch),
+                     dim3(256), 0, S(stream), (const float*)x,
+                     (const float*)p1_ws, (const float*)p2_ws,
+                     (const float*)d3_ws, (const float*)ga1_ws,
+                     (const float*)ga2_ws, (const float*)gh1_ws,
+                     (const float*)glog_ws, (float*)part_ws, B, bchunk,
+                     0);
+  GwPtrs gp{};
+  for (int i = 0; i < 8; ++i) gp.p[i] = (float*)grd_v[i];
+  unsigned long long* sb = (training && seed_dev)
+      ? (unsigned long long*)seed_dev : nullptr;
+  if (prm_v.empty()) {
+    hipLaunchKernelGGL(net_gw_combine_kernel,
+                       dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
+                       S(stream), (const float*)part_ws, gp, nch,
+                       (const float*)loss_part, (float*)loss_out,
+                       nblk, sb);
+  } else {
+    GwPtrs pp{}, bp{};
+    for (int i = 0; i < 8; ++i) {
+      pp.p[i] = (float*)prm_v[i];
+      bp.p[i] = (i < (int)buf_v.size()) ? (float*)buf_v[i] : nullptr;
+    }
+    hipLaunchKernelGGL(net_gw_combine_sgd_kernel,
+                       dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
+                       S(stream), (const float*)part_ws, gp, pp, bp, nch,
+                       (float)lr, (float)mu, (const float*)loss_part,
+                       (float*)loss_out, nblk, sb);
+  }
+}
+
 // net_fused_bwd + the optimizer update fused into the combine kernel
 // (single-GPU path: no all-reduce between combine and step).  The
 // segmented-partial weight-gradient path handles any batch size.
@@ -2458,6 +2591,7 @@ PYBIND11_MODULE(_kernels, m) {
   m.def("sgd_step", &sgd_step);
   m.def("net_fused_fwd", &net_fused_fwd);
   m.def("net_fused_bwd", &net_fused_bwd);
+  m.def("net_fused_fwdbwd", &net_fused_fwdbwd);
   m.def("net_step", &net_step);
   m.def("net_step_available", &net_step_available);
   m.def("net_gw_partial_raw", &net_gw_partial_raw);

@@ -196,6 +196,45 @@ def net_fused_step_opt(net, x: torch.Tensor, tgt: torch.Tensor,
     return ws["loss"]
 
 
+def net_fused_step_fb(net, x: torch.Tensor, tgt: torch.Tensor,
+                      opt=None) -> torch.Tensor:
+    """Three-dispatch training step: forward AND data-backward share
+    ONE kernel (each workgroup runs its sample's fwd then immediately
+    its bwd — valid because the loss gradient is the constant 1 in the
+    training loop), then gw-partial, then combine.  With ``opt`` (a
+    FusedSGD) the combine also applies the update; without it, grads
+    land in ``.grad`` for the DP all-reduce.  Trade-off vs
+    net_fused_step: one dispatch + w2 re-staging saved, but the
+    backward loses its sibling-workgroup split (grid = B), so at small
+    B the chip is underfilled — measured, see profiles/."""
+    k = load_native("_kernels")
+    B = x.shape[0]
+    ws = _ws(B, x.device)
+    params = [net.conv1.weight, net.conv1.bias, net.conv2.weight,
+              net.conv2.bias, net.fc1.weight, net.fc1.bias,
+              net.fc2.weight, net.fc2.bias]
+    for p in params:
+        if p.grad is None:
+            p.grad = torch.empty_like(p)
+    if opt is not None:
+        prm = [p.data_ptr() for p in params]
+        bufs = [b.data_ptr() for b in opt._bufs] if opt._bufs else []
+        lr, mu = opt.lr, opt.momentum
+    else:
+        prm, bufs, lr, mu = [], [], 0.0, 0.0
+    k.net_fused_fwdbwd(
+        x.data_ptr(), *[p.data_ptr() for p in params], tgt.data_ptr(),
+        ws["p1"].data_ptr(), ws["idx1"].data_ptr(), ws["m2"].data_ptr(),
+        ws["p2"].data_ptr(), ws["idx2"].data_ptr(), ws["h1"].data_ptr(),
+        ws["m3"].data_ptr(), ws["d3"].data_ptr(), ws["logp"].data_ptr(),
+        ws["glog"].data_ptr(), ws["gh1"].data_ptr(), ws["ga2"].data_ptr(),
+        ws["ga1"].data_ptr(), ws["part"].data_ptr(),
+        [p.grad.data_ptr() for p in params], prm, bufs, lr, mu, B,
+        net.training, ws["loss_part"].data_ptr(), ws["loss"].data_ptr(),
+        _seed_ptr(x.device), _stream())
+    return ws["loss"]
+
+
 def net_step_available() -> bool:
     """True when the device supports the single-launch cooperative
     training-step kernel (hipLaunchCooperativeKernel)."""

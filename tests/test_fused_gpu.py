@@ -238,3 +238,58 @@ def test_fused_step_tiny_odd_batch():
     assert torch.allclose(loss_g.cpu(), loss_c, atol=1e-5)
     for (n, pc), pg in zip(net_c.named_parameters(), net_g.parameters()):
         assert torch.allclose(pg.grad.cpu(), pc.grad, atol=2e-4), n
+
+
+def test_fwdbwd_matches_fused_eval():
+    """The combined fwd+bwd single-kernel step must produce identical
+    loss and grads to the separate fwd/bwd dispatch path (eval mode:
+    bit-deterministic)."""
+    from dist_tuto_pth_amd.ops.fused import net_fused_step_fb
+    net_c, net_g, x, tgt = _mk(11)
+    flat = attach_flat_grads(net_g)
+    loss_a = net_fused_step(net_g, x.to(DEV), tgt.to(DEV)).clone()
+    g_a = flat.clone()
+    loss_b = net_fused_step_fb(net_g, x.to(DEV), tgt.to(DEV)).clone()
+    torch.cuda.synchronize()
+    assert torch.allclose(loss_a, loss_b, atol=1e-6)
+    assert torch.equal(flat, g_a) or torch.allclose(flat, g_a, atol=1e-6)
+
+
+def test_fwdbwd_sgd_matches_separate_eval():
+    """fwdbwd with the fused SGD combine == fwd/bwd + separate SGD."""
+    from dist_tuto_pth_amd.optim import FusedSGD
+    from dist_tuto_pth_amd.ops.fused import net_fused_step_fb
+    net_c, net_a, x, tgt = _mk(12)
+    torch.manual_seed(12)
+    net_b = Net().eval().to(DEV)
+    net_b.load_state_dict(net_a.state_dict())
+    attach_flat_grads(net_a)
+    attach_flat_grads(net_b)
+    opt_a = FusedSGD(net_a.parameters(), lr=0.01, momentum=0.5)
+    opt_b = FusedSGD(net_b.parameters(), lr=0.01, momentum=0.5)
+    xg, tg = x.to(DEV), tgt.to(DEV)
+    for _ in range(3):
+        net_fused_step(net_a, xg, tg)
+        opt_a.step()
+        net_fused_step_fb(net_b, xg, tg, opt_b)
+    torch.cuda.synchronize()
+    for pa, pb in zip(net_a.parameters(), net_b.parameters()):
+        assert torch.allclose(pa, pb, atol=1e-6), \
+            (pa - pb).abs().max()
+
+
+def test_fwdbwd_training_convergence():
+    from dist_tuto_pth_amd.optim import FusedSGD
+    from dist_tuto_pth_amd.ops.fused import net_fused_step_fb
+    torch.manual_seed(3)
+    net = Net().to(DEV)
+    attach_flat_grads(net)
+    opt = FusedSGD(net.parameters(), lr=0.05, momentum=0.9)
+    g = torch.Generator().manual_seed(4)
+    x = torch.randn(64, 1, 28, 28, generator=g).to(DEV)
+    tgt = torch.randint(0, 10, (64,), generator=g).to(DEV)
+    losses = []
+    for _ in range(30):
+        loss = net_fused_step_fb(net, x, tgt, opt)
+        losses.append(float(loss))
+    assert losses[-1] < losses[0] * 0.5, (losses[0], losses[-1])
