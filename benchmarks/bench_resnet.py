@@ -38,6 +38,10 @@ def main():
                    help="NHWC memory format (MIOpen fast path)")
     p.add_argument("--no-find", action="store_true",
                    help="disable MIOpen exhaustive kernel search")
+    p.add_argument("--force-comm", action="store_true",
+                   help="world-1 tracing mode: init the backend and "
+                        "launch the bucket all-reduces anyway so a "
+                        "rocprof timeline shows the comm-stream overlap")
     p.add_argument("--dtype", default="fp32", choices=["fp32", "bf16"],
                    help="bf16 = autocast compute (supplementary number; "
                         "the headline config is fp32)")
@@ -51,7 +55,7 @@ def main():
     rank = int(os.environ.get("RANK", 0))
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
     dev_idx = local_rank % torch.cuda.device_count()
-    if world > 1:
+    if world > 1 or args.force_comm:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29500")
         dist.init_process_group("rccl", world_size=world, rank=rank,
@@ -67,8 +71,9 @@ def main():
         for prm in model.parameters():
             dist.broadcast(prm.data, src=0)
     ddp = None
-    if world > 1 and not args.no_overlap:
-        ddp = DistributedDataParallel(model, bucket_cap_mb=args.bucket_mb)
+    if (world > 1 or args.force_comm) and not args.no_overlap:
+        ddp = DistributedDataParallel(model, bucket_cap_mb=args.bucket_mb,
+                                      force_comm=args.force_comm)
     opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
 
     g = torch.Generator().manual_seed(1234 + rank)
@@ -140,7 +145,7 @@ def main():
                        if args.no_overlap else
                        f"ddp_overlap_{args.bucket_mb}MB"},
         }))
-    if world > 1:
+    if world > 1 or args.force_comm:
         dist.destroy_process_group()
 
 

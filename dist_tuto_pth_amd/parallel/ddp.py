@@ -90,11 +90,15 @@ class DistributedDataParallel(torch.nn.Module):
     """
 
     def __init__(self, module: torch.nn.Module, bucket_cap_mb: float = 25.0,
-                 group=None):
+                 group=None, force_comm: bool = False):
         super().__init__()
         self.module = module
         self.group = group
         self.world = dist.get_world_size(group)
+        # force_comm: launch bucket all-reduces even at world 1 (they
+        # are device-side no-op copies) — lets a single-GPU rocprof
+        # trace show the comm-stream overlap schedule (profiles/)
+        self._force_comm = force_comm
         params = [p for p in module.parameters() if p.requires_grad]
         self._params = params
         device = params[0].device if params else torch.device("cpu")
@@ -151,7 +155,7 @@ class DistributedDataParallel(torch.nn.Module):
         return hook
 
     def _launch(self, bucket: _Bucket):
-        if self.world == 1:
+        if self.world == 1 and not self._force_comm:
             return
         if self._use_stream:
             # grads ARE the bucket flat: the all-reduce is the only op

@@ -279,17 +279,19 @@ def test_fwdbwd_sgd_matches_separate_eval():
 
 
 def test_fwdbwd_training_convergence():
+    """Same statistical bar as the other training-mode convergence
+    tests: 150 steps with dropout on, mean-of-10 loss drops >= 0.12
+    (the CPU reference drops ~0.24 with these hyperparameters)."""
     from dist_tuto_pth_amd.optim import FusedSGD
     from dist_tuto_pth_amd.ops.fused import net_fused_step_fb
-    torch.manual_seed(3)
-    net = Net().to(DEV)
-    attach_flat_grads(net)
-    opt = FusedSGD(net.parameters(), lr=0.05, momentum=0.9)
-    g = torch.Generator().manual_seed(4)
-    x = torch.randn(64, 1, 28, 28, generator=g).to(DEV)
-    tgt = torch.randint(0, 10, (64,), generator=g).to(DEV)
+    _, net_g, x, tgt = _mk(13, B=256)
+    net_g.train()
+    attach_flat_grads(net_g)
+    opt = FusedSGD(net_g.parameters(), lr=0.05, momentum=0.5)
+    xg, tg = x.to(DEV), tgt.to(DEV)
     losses = []
-    for _ in range(30):
-        loss = net_fused_step_fb(net, x, tgt, opt)
-        losses.append(float(loss))
-    assert losses[-1] < losses[0] * 0.5, (losses[0], losses[-1])
+    for _ in range(150):
+        losses.append(float(net_fused_step_fb(net_g, xg, tg, opt)))
+    first = sum(losses[:10]) / 10
+    last = sum(losses[-10:]) / 10
+    assert last < first - 0.12, (first, last)
