@@ -42,12 +42,11 @@ def main():
         top = sorted(names.items(), key=lambda kv: -kv[1])[:6]
         lines.append(f"stream {st}: {len(ks)} kernels, busy {busy:.2f} ms"
                      f"; top: {top}")
-    # overlap: comm stream = the non-main stream with the most kernels
+    # overlap of EVERY side stream against the main (compute) stream
     main_st = max(streams, key=lambda st: len(streams[st]))
-    others = [st for st in streams if st != main_st]
-    if others:
-        comm_st = max(others, key=lambda st: len(streams[st]))
-        mains = streams[main_st]
+    mains = streams[main_st]
+    t0 = rows[0][0]
+    for comm_st in sorted(st for st in streams if st != main_st):
         tot, ovl = 0, 0
         per = []
         for s, e, name in streams[comm_st]:
@@ -63,15 +62,15 @@ def main():
             per.append((s, e, name.split('(')[0][:40],
                         100.0 * o / max(e - s, 1)))
         lines.append(
-            f"comm stream {comm_st}: {tot/1e6:.2f} ms total, "
-            f"{ovl/1e6:.2f} ms overlapped with stream {main_st} compute "
+            f"stream {comm_st} vs main {main_st}: {tot/1e6:.3f} ms "
+            f"total, {ovl/1e6:.3f} ms overlapped "
             f"({100.0*ovl/max(tot,1):.1f}%)")
-        t0 = rows[0][0]
-        lines.append("last 30 comm-stream kernels "
-                     "(start us, dur us, %overlapped):")
-        for s, e, n, pct in per[-30:]:
-            lines.append(f"  {(s-t0)/1e3:12.1f} {(e-s)/1e3:9.1f} "
-                         f"{pct:5.1f}%  {n}")
+        if len(per) >= 10:
+            lines.append(f"  last 15 kernels on stream {comm_st} "
+                         "(start us, dur us, %overlapped):")
+            for s, e, n, pct in per[-15:]:
+                lines.append(f"  {(s-t0)/1e3:12.1f} {(e-s)/1e3:9.1f} "
+                             f"{pct:5.1f}%  {n}")
     text = "\n".join(lines) + "\n"
     if out_path:
         with open(out_path, "w") as f:
