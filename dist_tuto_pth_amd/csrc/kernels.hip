@@ -2037,6 +2037,20 @@ void sgd_step(const std::vector<uintptr_t>& ps,
   }
 }
 
+
+// gw chunk count: default 32 (or B when smaller); DTP_GW_NCH overrides.
+static int gw_nch(int B) {
+  static int env_nch = -2;
+  if (env_nch == -2) {
+    const char* e = std::getenv("DTP_GW_NCH");
+    env_nch = e ? std::atoi(e) : -1;
+  }
+  int nch = (env_nch > 0) ? env_nch : 32;
+  if (nch > 32) nch = 32;  // the part workspace holds 32 rows (_ws)
+  if (nch > B) nch = B;
+  return nch;
+}
+
 void net_fused_fwd(uintptr_t x, uintptr_t w1, uintptr_t b1, uintptr_t w2,
                    uintptr_t b2, uintptr_t wf1, uintptr_t bf1,
                    uintptr_t wf2, uintptr_t bf2, uintptr_t tgt,
@@ -2160,9 +2174,11 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
     return;
   }
   // one segmented partial kernel over <=32 batch chunks + one combine
-  // (no memsets, no atomics); any batch size
-  const int bchunk = (B + 31) / 32;
-  const int nch = (B + bchunk - 1) / bchunk;
+  // (no memsets, no atomics); any batch size.  DTP_GW_NCH overrides
+  // the chunk count (fewer chunks = less partial traffic for the
+  // combine, more batch per partial block — sweep on hardware).
+  const int nch = gw_nch(B);
+  const int bchunk = (B + nch - 1) / nch;
   hipLaunchKernelGGL(net_gw_partial_kernel, dim3(GW_TILES, nch),
                      dim3(256), 0, S(stream), (const float*)x,
                      (const float*)p1_ws, (const float*)p2_ws,
@@ -2218,8 +2234,8 @@ void net_fused_fwdbwd(
                      (float*)loss_part,
                      (const unsigned long long*)seed_dev, B,
                      training ? 1 : 0);
-  const int bchunk = (B + 31) / 32;
-  const int nch = (B + bchunk - 1) / bchunk;
+  const int nch = gw_nch(B);
+  const int bchunk = (B + nch - 1) / nch;
   hipLaunchKernelGGL(net_gw_partial_kernel, dim3(GW_TILES, nch),
                      dim3(256), 0, S(stream), (const float*)x,
                      (const float*)p1_ws, (const float*)p2_ws,
@@ -2287,8 +2303,8 @@ void net_fused_bwd_sgd(uintptr_t x, uintptr_t w2, uintptr_t wf1,
                      (const float*)logp_ws, (float*)glog_ws,
                      (float*)gh1_ws, (float*)ga2_ws, (float*)ga1_ws, B,
                      training ? 1 : 0, split);
-  const int bchunk = (B + 31) / 32;
-  const int nch = (B + bchunk - 1) / bchunk;
+  const int nch = gw_nch(B);
+  const int bchunk = (B + nch - 1) / nch;
   hipLaunchKernelGGL(net_gw_partial_kernel, dim3(GW_TILES, nch),
                      dim3(256), 0, S(stream), (const float*)x,
                      (const float*)p1_ws, (const float*)p2_ws,
