@@ -1356,27 +1356,40 @@ __device__ __forceinline__ void net_gw_tile(
     const float* __restrict__ glog_ws) {
   if (tile < T_CONV2) {  // conv2: gw [20][10][5][5] + gb [20]
     // one block per output channel k: stage the batch element's full
-    // p1 plane (1440 floats) and its ga2 row (64) through LDS once,
-    // then the 250 weight threads reduce from LDS (each p1 value was
-    // being re-read ~11x from L1 in the element-per-thread form).
-    // Four accumulator chains over output-row parity.
+    // p1 plane (1440 floats) and its ga2 row (64) through LDS, then
+    // the 250 weight threads reduce from LDS (each p1 value was being
+    // re-read ~11x from L1 in the element-per-thread form).  Staging
+    // is DOUBLE-BUFFERED: while the 250 compute lanes fold sample b
+    // from buffer p, all 256 lanes' global loads for b+1 land in
+    // buffer p^1 — one barrier per sample instead of two, and the
+    // global-load latency hides under the 80-FMA fold (r2; the
+    // single-buffer form is in the r1 history).
     const int k = tile;
-    __shared__ float sp1[N_P1];
-    __shared__ float sg2[64];
+    __shared__ float sp1[2][N_P1];
+    __shared__ float sg2[2][64];
     const int c = tid / 25, r = (tid / 5) % 5, sx = tid % 5;
     float q0 = 0.f, q1 = 0.f, q2 = 0.f, q3 = 0.f;
     float be = 0.f, bo = 0.f;
+    // prologue: stage b0 into buffer 0
+    for (int i = tid; i < N_P1; i += 256)
+      sp1[0][i] = p1_ws[(int64_t)b0 * N_P1 + i];
+    if (tid < 64) sg2[0][tid] = ga2_ws[(int64_t)b0 * N_A2 + k * 64 + tid];
+    __syncthreads();
     for (int b = b0; b < b1; ++b) {
-      __syncthreads();
-      for (int i = tid; i < N_P1; i += 256)
-        sp1[i] = p1_ws[(int64_t)b * N_P1 + i];
-      if (tid < 64) sg2[tid] = ga2_ws[(int64_t)b * N_A2 + k * 64 + tid];
-      __syncthreads();
+      const int cur = (b - b0) & 1;
+      if (b + 1 < b1) {  // stage next sample into the other buffer
+        for (int i = tid; i < N_P1; i += 256)
+          sp1[cur ^ 1][i] = p1_ws[(int64_t)(b + 1) * N_P1 + i];
+        if (tid < 64)
+          sg2[cur ^ 1][tid] =
+              ga2_ws[(int64_t)(b + 1) * N_A2 + k * 64 + tid];
+      }
       if (tid < 250) {
-        const float* xc = sp1 + c * 144 + r * 12 + sx;
+        const float* xc = sp1[cur] + c * 144 + r * 12 + sx;
+        const float* sg = sg2[cur];
         #pragma unroll
         for (int oh = 0; oh < 8; ++oh) {
-          const float* gr = sg2 + oh * 8;
+          const float* gr = sg + oh * 8;
           const float* xr = xc + oh * 12;
           const float s0 = gr[0] * xr[0] + gr[1] * xr[1] +
                            gr[2] * xr[2] + gr[3] * xr[3];
@@ -1388,12 +1401,14 @@ __device__ __forceinline__ void net_gw_tile(
           else q3 += s0 + s1;
         }
       } else if (tid == 250) {  // bias: two chains over the 64 values
+        const float* sg = sg2[cur];
         #pragma unroll
         for (int j = 0; j < 64; j += 2) {
-          be += sg2[j];
-          bo += sg2[j + 1];
+          be += sg[j];
+          bo += sg[j + 1];
         }
       }
+      __syncthreads();
     }
     if (tid < 250) my[OFF_W2 + k * 250 + tid] = (q0 + q1) + (q2 + q3);
     if (tid == 250) my[OFF_B2 + k] = be + bo;
@@ -1436,39 +1451,71 @@ __device__ __forceinline__ void net_gw_tile(
     // over 8 sub-blocks (3 output rows each) so the grid column's
     // straggler block shrinks 8x.  Sub 0 writes the canonical region,
     // subs 1-7 the extension rows summed by the combine kernel.
+    // r2: the sample's ga1 band (10 k x 3 rows) and the 7 x-rows it
+    // touches are staged through a DOUBLE-BUFFERED LDS tile — each
+    // ga1 row was being re-read ~25x and each x row ~15x from L1 by
+    // the item loop; now they are read once from HBM per sample and
+    // the next sample's loads hide under the fold.
     const int sub = tile;
     const int oh0 = sub * 3;
     __shared__ float wacc[260];
+    __shared__ float sga[2][30 * 24];   // [k*3 + oh_j][ow]
+    __shared__ float sxr[2][7 * 28];    // x rows oh0 .. oh0+6
     for (int i = tid; i < 260; i += 256) wacc[i] = 0.f;
+    // prologue: stage b0 into buffer 0
+    for (int i = tid; i < 30 * 24; i += 256) {
+      const int kk = i / 72, rem = i % 72;
+      sga[0][i] = ga1_ws[(int64_t)b0 * N_A1 + kk * 576 +
+                         (oh0 + rem / 24) * 24 + rem % 24];
+    }
+    for (int i = tid; i < 7 * 28; i += 256)
+      sxr[0][i] = x[(int64_t)b0 * 784 + (oh0 + i / 28) * 28 + i % 28];
     __syncthreads();
-    for (int it = tid; it < 750 + 30; it += 256) {
-      float a = 0.f;
-      if (it < 750) {
-        const int e = it / 3, oh = oh0 + it % 3;
-        const int k = e / 25, r = (e / 5) % 5, sx = e % 5;
-        for (int b = b0; b < b1; ++b) {
-          const float* grow = ga1_ws + (int64_t)b * N_A1 + k * 576 +
-                              oh * 24;
-          const float* xrow = x + (int64_t)b * 784 + (oh + r) * 28 + sx;
+    float accs[4] = {0.f, 0.f, 0.f, 0.f};  // ceil(780/256) items/thread
+    for (int b = b0; b < b1; ++b) {
+      const int cur = (b - b0) & 1;
+      if (b + 1 < b1) {
+        for (int i = tid; i < 30 * 24; i += 256) {
+          const int kk = i / 72, rem = i % 72;
+          sga[cur ^ 1][i] = ga1_ws[(int64_t)(b + 1) * N_A1 + kk * 576 +
+                                   (oh0 + rem / 24) * 24 + rem % 24];
+        }
+        for (int i = tid; i < 7 * 28; i += 256)
+          sxr[cur ^ 1][i] =
+              x[(int64_t)(b + 1) * 784 + (oh0 + i / 28) * 28 + i % 28];
+      }
+      int slot = 0;
+      for (int it = tid; it < 750 + 30; it += 256, ++slot) {
+        float a = 0.f;
+        if (it < 750) {
+          const int e = it / 3, oh_j = it % 3;
+          const int r = (e / 5) % 5, sx = e % 5;
+          const float* grow = sga[cur] + (e / 25 * 3 + oh_j) * 24;
+          const float* xrow = sxr[cur] + (oh_j + r) * 28 + sx;
           float ae = 0.f, ao = 0.f;
           #pragma unroll
           for (int ow = 0; ow < 24; ow += 2) {
             ae += grow[ow] * xrow[ow];
             ao += grow[ow + 1] * xrow[ow + 1];
           }
-          a += ae + ao;
-        }
-        atomicAdd(&wacc[e], a);
-      } else {
-        const int j = it - 750;
-        const int k = j / 3, oh = oh0 + j % 3;
-        for (int b = b0; b < b1; ++b) {
-          const float* grow = ga1_ws + (int64_t)b * N_A1 + k * 576 +
-                              oh * 24;
+          a = ae + ao;
+        } else {
+          const int j = it - 750;
+          const float* grow = sga[cur] + (j / 3 * 3 + j % 3) * 24;
           #pragma unroll 8
           for (int ow = 0; ow < 24; ++ow) a += grow[ow];
         }
-        atomicAdd(&wacc[250 + k], a);
+        accs[slot] += a;
+      }
+      __syncthreads();
+    }
+    {
+      int slot = 0;
+      for (int it = tid; it < 750 + 30; it += 256, ++slot) {
+        if (it < 750)
+          atomicAdd(&wacc[it / 3], accs[slot]);
+        else
+          atomicAdd(&wacc[250 + (it - 750) / 3], accs[slot]);
       }
     }
     __syncthreads();
