@@ -1451,71 +1451,45 @@ __device__ __forceinline__ void net_gw_tile(
     // over 8 sub-blocks (3 output rows each) so the grid column's
     // straggler block shrinks 8x.  Sub 0 writes the canonical region,
     // subs 1-7 the extension rows summed by the combine kernel.
-    // r2: the sample's ga1 band (10 k x 3 rows) and the 7 x-rows it
-    // touches are staged through a DOUBLE-BUFFERED LDS tile — each
-    // ga1 row was being re-read ~25x and each x row ~15x from L1 by
-    // the item loop; now they are read once from HBM per sample and
-    // the next sample's loads hide under the fold.
+    // (r2 note: an LDS-staged double-buffered variant of this tile was
+    // MEASURED SLOWER - 8.7 -> 11.3 us at B=128, 281 -> 325 us at
+    // B=4096: the ~25x grow-row re-reads are L1 hits and the staging
+    // address math + extra barrier cost more than they save.  This
+    // direct-from-L1 form stands; the conv2 tile, whose re-reads miss
+    // L1, DID win from the same treatment.)
     const int sub = tile;
     const int oh0 = sub * 3;
     __shared__ float wacc[260];
-    __shared__ float sga[2][30 * 24];   // [k*3 + oh_j][ow]
-    __shared__ float sxr[2][7 * 28];    // x rows oh0 .. oh0+6
     for (int i = tid; i < 260; i += 256) wacc[i] = 0.f;
-    // prologue: stage b0 into buffer 0
-    for (int i = tid; i < 30 * 24; i += 256) {
-      const int kk = i / 72, rem = i % 72;
-      sga[0][i] = ga1_ws[(int64_t)b0 * N_A1 + kk * 576 +
-                         (oh0 + rem / 24) * 24 + rem % 24];
-    }
-    for (int i = tid; i < 7 * 28; i += 256)
-      sxr[0][i] = x[(int64_t)b0 * 784 + (oh0 + i / 28) * 28 + i % 28];
     __syncthreads();
-    float accs[4] = {0.f, 0.f, 0.f, 0.f};  // ceil(780/256) items/thread
-    for (int b = b0; b < b1; ++b) {
-      const int cur = (b - b0) & 1;
-      if (b + 1 < b1) {
-        for (int i = tid; i < 30 * 24; i += 256) {
-          const int kk = i / 72, rem = i % 72;
-          sga[cur ^ 1][i] = ga1_ws[(int64_t)(b + 1) * N_A1 + kk * 576 +
-                                   (oh0 + rem / 24) * 24 + rem % 24];
-        }
-        for (int i = tid; i < 7 * 28; i += 256)
-          sxr[cur ^ 1][i] =
-              x[(int64_t)(b + 1) * 784 + (oh0 + i / 28) * 28 + i % 28];
-      }
-      int slot = 0;
-      for (int it = tid; it < 750 + 30; it += 256, ++slot) {
-        float a = 0.f;
-        if (it < 750) {
-          const int e = it / 3, oh_j = it % 3;
-          const int r = (e / 5) % 5, sx = e % 5;
-          const float* grow = sga[cur] + (e / 25 * 3 + oh_j) * 24;
-          const float* xrow = sxr[cur] + (oh_j + r) * 28 + sx;
+    for (int it = tid; it < 750 + 30; it += 256) {
+      float a = 0.f;
+      if (it < 750) {
+        const int e = it / 3, oh = oh0 + it % 3;
+        const int k = e / 25, r = (e / 5) % 5, sx = e % 5;
+        for (int b = b0; b < b1; ++b) {
+          const float* grow = ga1_ws + (int64_t)b * N_A1 + k * 576 +
+                              oh * 24;
+          const float* xrow = x + (int64_t)b * 784 + (oh + r) * 28 + sx;
           float ae = 0.f, ao = 0.f;
           #pragma unroll
           for (int ow = 0; ow < 24; ow += 2) {
             ae += grow[ow] * xrow[ow];
             ao += grow[ow + 1] * xrow[ow + 1];
           }
-          a = ae + ao;
-        } else {
-          const int j = it - 750;
-          const float* grow = sga[cur] + (j / 3 * 3 + j % 3) * 24;
+          a += ae + ao;
+        }
+        atomicAdd(&wacc[e], a);
+      } else {
+        const int j = it - 750;
+        const int k = j / 3, oh = oh0 + j % 3;
+        for (int b = b0; b < b1; ++b) {
+          const float* grow = ga1_ws + (int64_t)b * N_A1 + k * 576 +
+                              oh * 24;
           #pragma unroll 8
           for (int ow = 0; ow < 24; ++ow) a += grow[ow];
         }
-        accs[slot] += a;
-      }
-      __syncthreads();
-    }
-    {
-      int slot = 0;
-      for (int it = tid; it < 750 + 30; it += 256, ++slot) {
-        if (it < 750)
-          atomicAdd(&wacc[it / 3], accs[slot]);
-        else
-          atomicAdd(&wacc[250 + (it - 750) / 3], accs[slot]);
+        atomicAdd(&wacc[250 + k], a);
       }
     }
     __syncthreads();
