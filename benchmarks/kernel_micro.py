@@ -111,9 +111,9 @@ def main():
         results[f"bwd+gw (split={sp})"] = time_fn(bwd)
     os.environ.pop("DTP_BWD_SPLIT", None)
     results["gw conv2 tiles"] = time_fn(seg(0, T_CONV2))
-    results["gw fc1 tiles"] = time_fn(seg(T_CONV2, T_FC1))
-    results["gw conv1 tiles"] = time_fn(seg(T_CONV2 + T_FC1, T_CONV1))
-    results["gw fc2 tiles"] = time_fn(seg(T_CONV2 + T_FC1 + T_CONV1, T_FC2))
+    results["gw conv1 tiles"] = time_fn(seg(T_CONV2, T_CONV1))
+    results["gw fc1 tiles"] = time_fn(seg(T_CONV2 + T_CONV1, T_FC1))
+    results["gw fc2 tiles"] = time_fn(seg(T_CONV2 + T_CONV1 + T_FC1, T_FC2))
     results["gw all tiles"] = time_fn(seg(0, T_CONV2 + T_FC1 + T_CONV1 + T_FC2))
     results["gw combine"] = time_fn(lambda: k.net_gw_combine_raw(
         ws["part"].data_ptr(), [p.grad.data_ptr() for p in params], nch,

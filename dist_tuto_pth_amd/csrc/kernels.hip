@@ -1328,9 +1328,9 @@ net_fused_fwdbwd_kernel(
 #define OFF_BF1 21280     // 50
 #define OFF_WF2 21330     // 500
 #define OFF_BF2 21830     // 10
-#define T_CONV2 20        // one block per conv2 output channel
-static_assert(T_CONV2 == N_C2K, "conv2 gw tiling is one block per "
-              "output channel");
+#define T_CONV2 20        // conv2 tile columns (pair mode uses 10)
+static_assert(T_CONV2 == N_C2K, "conv2 gw tiling: one tile per output "
+              "channel (single mode) or per channel pair (pair mode)");
 #define T_FC1 63          // ceil(16050/256)
 #define T_CONV1 8         // 8 sub-blocks over 3-output-row bands
 #define T_FC2 2           // ceil(510/256)
@@ -1339,6 +1339,120 @@ static_assert(T_CONV2 == N_C2K, "conv2 gw tiling is one block per "
 // canonical [OFF_W1,OFF_B1] region, subs 1-7 to an extension past
 // GW_TOTAL; the combine kernel folds the extension back in.
 #define GW_ROW (GW_TOTAL + 7 * 260)
+
+// conv2 weight-gradient fold, templated on channels-per-block so every
+// accumulator index is compile-time (a runtime nk spilled q[][] to
+// scratch and tripled the kernel — r2 ledger).  Thread = (ohGroup, c,
+// r), 200 lanes, owns all FIVE sx weights of its (c, r) row: one
+// 12-value xr window + one 8-value gr row feed 40 FMA per channel.
+// Staging is double-buffered float4.
+template <int NK>
+__device__ __forceinline__ void net_gw_conv2_fold(
+    int k0, int tid, int b0, int b1, float* __restrict__ my,
+    const float* __restrict__ p1_ws, const float* __restrict__ ga2_ws,
+    float* __restrict__ sp1_lds,    // 2*N_P1 floats, 16 B aligned
+    float* __restrict__ sg2_lds,    // 2*128 floats, 16 B aligned
+    float* __restrict__ wacc_lds) { // 2*251 floats
+  // LDS is provided by the caller so the NK=1 and NK=2 instantiations
+  // share ONE static allocation (in-template statics doubled the
+  // kernel's LDS and cost occupancy — r2 ledger)
+  float* sp1[2] = {sp1_lds, sp1_lds + N_P1};
+  float* sg2[2] = {sg2_lds, sg2_lds + 64 * NK};
+  float* wacc[NK];
+  #pragma unroll
+  for (int kk = 0; kk < NK; ++kk) wacc[kk] = wacc_lds + kk * 251;
+  const int ohp = tid / 50;          // 0..3 (tid < 200)
+  const int cr = tid % 50;
+  const int c = cr / 5, r = cr % 5;
+  float q[NK][5] = {};
+  float be[NK] = {}, bo[NK] = {};
+  for (int i = tid; i < NK * 251; i += 256) wacc[i / 251][i % 251] = 0.f;
+  // prologue: stage b0 into buffer 0 (sp1 rows and the per-sample
+  // bases are 16 B aligned)
+  {
+    float4* d4 = reinterpret_cast<float4*>(sp1[0]);
+    const float4* s4 = reinterpret_cast<const float4*>(
+        p1_ws + (int64_t)b0 * N_P1);
+    for (int i = tid; i < N_P1 / 4; i += 256) d4[i] = s4[i];
+    float4* g4 = reinterpret_cast<float4*>(sg2[0]);
+    const float4* a4 = reinterpret_cast<const float4*>(
+        ga2_ws + (int64_t)b0 * N_A2 + k0 * 64);
+    if (tid < 16 * NK) g4[tid] = a4[tid];
+  }
+  __syncthreads();
+  for (int b = b0; b < b1; ++b) {
+    const int cur = (b - b0) & 1;
+    if (b + 1 < b1) {  // stage next sample into the other buffer
+      float4* d4 = reinterpret_cast<float4*>(sp1[cur ^ 1]);
+      const float4* s4 = reinterpret_cast<const float4*>(
+          p1_ws + (int64_t)(b + 1) * N_P1);
+      for (int i = tid; i < N_P1 / 4; i += 256) d4[i] = s4[i];
+      float4* g4 = reinterpret_cast<float4*>(sg2[cur ^ 1]);
+      const float4* a4 = reinterpret_cast<const float4*>(
+          ga2_ws + (int64_t)(b + 1) * N_A2 + k0 * 64);
+      if (tid < 16 * NK) g4[tid] = a4[tid];
+    }
+    if (tid < 200) {
+      #pragma unroll
+      for (int ohh = 0; ohh < 2; ++ohh) {
+        const int oh = ohp * 2 + ohh;
+        const float* xr = sp1[cur] + c * 144 + (r + oh) * 12;
+        float xv[12];
+        #pragma unroll
+        for (int j = 0; j < 12; ++j) xv[j] = xr[j];
+        #pragma unroll
+        for (int kk = 0; kk < NK; ++kk) {
+          const float* gr = sg2[cur] + kk * 64 + oh * 8;
+          #pragma unroll
+          for (int ow = 0; ow < 8; ++ow) {
+            const float g = gr[ow];
+            q[kk][0] += g * xv[ow];
+            q[kk][1] += g * xv[ow + 1];
+            q[kk][2] += g * xv[ow + 2];
+            q[kk][3] += g * xv[ow + 3];
+            q[kk][4] += g * xv[ow + 4];
+          }
+        }
+      }
+    } else if (tid == 200) {  // biases: two chains per channel
+      #pragma unroll
+      for (int kk = 0; kk < NK; ++kk) {
+        const float* sg = sg2[cur] + kk * 64;
+        #pragma unroll
+        for (int j = 0; j < 64; j += 2) {
+          be[kk] += sg[j];
+          bo[kk] += sg[j + 1];
+        }
+      }
+    }
+    __syncthreads();
+  }
+  if (tid < 200) {
+    const int e = c * 25 + r * 5;
+    #pragma unroll
+    for (int kk = 0; kk < NK; ++kk) {
+      atomicAdd(&wacc[kk][e + 0], q[kk][0]);
+      atomicAdd(&wacc[kk][e + 1], q[kk][1]);
+      atomicAdd(&wacc[kk][e + 2], q[kk][2]);
+      atomicAdd(&wacc[kk][e + 3], q[kk][3]);
+      atomicAdd(&wacc[kk][e + 4], q[kk][4]);
+    }
+  } else if (tid == 200) {
+    #pragma unroll
+    for (int kk = 0; kk < NK; ++kk) wacc[kk][250] = be[kk] + bo[kk];
+  }
+  __syncthreads();
+  if (tid < 250) {
+    #pragma unroll
+    for (int kk = 0; kk < NK; ++kk)
+      my[OFF_W2 + (k0 + kk) * 250 + tid] = wacc[kk][tid];
+  }
+  if (tid == 250) {
+    #pragma unroll
+    for (int kk = 0; kk < NK; ++kk)
+      my[OFF_B2 + k0 + kk] = wacc[kk][250];
+  }
+}
 
 // One (tile, batch-chunk) partial weight-gradient reduction.  Tiles
 // walk [conv2 | fc1 | conv1x4 | fc2]; partials land in my[GW_ROW]
@@ -1355,95 +1469,28 @@ __device__ __forceinline__ void net_gw_tile(
     const float* __restrict__ gh1_ws,
     const float* __restrict__ glog_ws) {
   if (tile < T_CONV2) {  // conv2: gw [20][10][5][5] + gb [20]
-    // one block per output channel k: stage the batch element's full
-    // p1 plane (1440 floats) and its ga2 row (64) through LDS, then
-    // the 250 weight threads reduce from LDS (each p1 value was being
-    // re-read ~11x from L1 in the element-per-thread form).  Staging
-    // is DOUBLE-BUFFERED: while the 250 compute lanes fold sample b
-    // from buffer p, all 256 lanes' global loads for b+1 land in
-    // buffer p^1 — one barrier per sample instead of two, and the
-    // global-load latency hides under the 80-FMA fold (r2; the
-    // single-buffer form is in the r1 history).
-    const int k = tile;
-    __shared__ float sp1[2][N_P1];
-    __shared__ float sg2[2][64];
-    const int c = tid / 25, r = (tid / 5) % 5, sx = tid % 5;
-    float q0 = 0.f, q1 = 0.f, q2 = 0.f, q3 = 0.f;
-    float be = 0.f, bo = 0.f;
-    // prologue: stage b0 into buffer 0
-    for (int i = tid; i < N_P1; i += 256)
-      sp1[0][i] = p1_ws[(int64_t)b0 * N_P1 + i];
-    if (tid < 64) sg2[0][tid] = ga2_ws[(int64_t)b0 * N_A2 + k * 64 + tid];
-    __syncthreads();
-    for (int b = b0; b < b1; ++b) {
-      const int cur = (b - b0) & 1;
-      if (b + 1 < b1) {  // stage next sample into the other buffer
-        for (int i = tid; i < N_P1; i += 256)
-          sp1[cur ^ 1][i] = p1_ws[(int64_t)(b + 1) * N_P1 + i];
-        if (tid < 64)
-          sg2[cur ^ 1][tid] =
-              ga2_ws[(int64_t)(b + 1) * N_A2 + k * 64 + tid];
-      }
-      if (tid < 250) {
-        const float* xc = sp1[cur] + c * 144 + r * 12 + sx;
-        const float* sg = sg2[cur];
-        #pragma unroll
-        for (int oh = 0; oh < 8; ++oh) {
-          const float* gr = sg + oh * 8;
-          const float* xr = xc + oh * 12;
-          const float s0 = gr[0] * xr[0] + gr[1] * xr[1] +
-                           gr[2] * xr[2] + gr[3] * xr[3];
-          const float s1 = gr[4] * xr[4] + gr[5] * xr[5] +
-                           gr[6] * xr[6] + gr[7] * xr[7];
-          if ((oh & 3) == 0) q0 += s0 + s1;
-          else if ((oh & 3) == 1) q1 += s0 + s1;
-          else if ((oh & 3) == 2) q2 += s0 + s1;
-          else q3 += s0 + s1;
-        }
-      } else if (tid == 250) {  // bias: two chains over the 64 values
-        const float* sg = sg2[cur];
-        #pragma unroll
-        for (int j = 0; j < 64; j += 2) {
-          be += sg[j];
-          bo += sg[j + 1];
-        }
-      }
-      __syncthreads();
+    // TWO block shapes, chosen by chunk size (measured, r2 ledger):
+    //  * pair mode (bchunk <= 16, i.e. B <= 512): one block folds TWO
+    //    output channels from one staging of the p1 plane — halves
+    //    the dominant HBM re-read; tiles 10..19 idle (the grid is
+    //    launch-granularity-bound here, idle blocks are free).
+    //  * single mode (large B): one channel per block — smaller
+    //    blocks pack the 2600-block grid without straggler tails,
+    //    which beats the re-read saving end to end.
+    __shared__ __attribute__((aligned(16))) float sp1_lds[2 * N_P1];
+    __shared__ __attribute__((aligned(16))) float sg2_lds[2 * 128];
+    __shared__ float wacc_lds[2 * 251];
+    if (b1 - b0 <= 16) {
+      if (tile < N_C2K / 2)
+        net_gw_conv2_fold<2>(tile * 2, tid, b0, b1, my, p1_ws, ga2_ws,
+                             sp1_lds, sg2_lds, wacc_lds);
+    } else {
+      net_gw_conv2_fold<1>(tile, tid, b0, b1, my, p1_ws, ga2_ws,
+                           sp1_lds, sg2_lds, wacc_lds);
     }
-    if (tid < 250) my[OFF_W2 + k * 250 + tid] = (q0 + q1) + (q2 + q3);
-    if (tid == 250) my[OFF_B2 + k] = be + bo;
     return;
   }
   tile -= T_CONV2;
-  if (tile < T_FC1) {  // fc1: gw [50][320] + gb [50]
-    const int i = tile * 256 + tid;
-    if (i < 16050) {
-      float acc = 0.f;
-      if (i < 16000) {
-        const int n = i / N_P2, k = i % N_P2;
-        float ae = 0.f, ao = 0.f;
-        int b = b0;
-        for (; b + 1 < b1; b += 2) {
-          ae += gh1_ws[(int64_t)b * N_H1 + n] *
-                p2_ws[(int64_t)b * N_P2 + k];
-          ao += gh1_ws[(int64_t)(b + 1) * N_H1 + n] *
-                p2_ws[(int64_t)(b + 1) * N_P2 + k];
-        }
-        if (b < b1)
-          ae += gh1_ws[(int64_t)b * N_H1 + n] *
-                p2_ws[(int64_t)b * N_P2 + k];
-        acc = ae + ao;
-        my[OFF_WF1 + i] = acc;
-      } else {
-        const int n = i - 16000;
-        for (int b = b0; b < b1; ++b)
-          acc += gh1_ws[(int64_t)b * N_H1 + n];
-        my[OFF_BF1 + n] = acc;
-      }
-    }
-    return;
-  }
-  tile -= T_FC1;
   if (tile < T_CONV1) {  // conv1: gw [10][1][5][5] + gb [10]
     // 250 outputs is too little parallelism for element-per-thread at
     // this cost (24x24 window x batch): split each output over its 24
@@ -1498,6 +1545,35 @@ __device__ __forceinline__ void net_gw_tile(
     return;
   }
   tile -= T_CONV1;
+  if (tile < T_FC1) {  // fc1: gw [50][320] + gb [50]
+    const int i = tile * 256 + tid;
+    if (i < 16050) {
+      float acc = 0.f;
+      if (i < 16000) {
+        const int n = i / N_P2, k = i % N_P2;
+        float ae = 0.f, ao = 0.f;
+        int b = b0;
+        for (; b + 1 < b1; b += 2) {
+          ae += gh1_ws[(int64_t)b * N_H1 + n] *
+                p2_ws[(int64_t)b * N_P2 + k];
+          ao += gh1_ws[(int64_t)(b + 1) * N_H1 + n] *
+                p2_ws[(int64_t)(b + 1) * N_P2 + k];
+        }
+        if (b < b1)
+          ae += gh1_ws[(int64_t)b * N_H1 + n] *
+                p2_ws[(int64_t)b * N_P2 + k];
+        acc = ae + ao;
+        my[OFF_WF1 + i] = acc;
+      } else {
+        const int n = i - 16000;
+        for (int b = b0; b < b1; ++b)
+          acc += gh1_ws[(int64_t)b * N_H1 + n];
+        my[OFF_BF1 + n] = acc;
+      }
+    }
+    return;
+  }
+  tile -= T_FC1;
   {  // fc2: gw [10][50] + gb [10]
     const int i = tile * 256 + tid;
     if (i < 510) {

@@ -84,15 +84,17 @@ def test_fused_step_matches_modular_gpu_eval():
 
 def test_fused_training_mode_statistics():
     """Train mode: dropout active; loss finite, grads nonzero, and the
-    dropout masks advance across steps (device-side seed bump)."""
+    dropout masks advance across STEPS.  Seed convention is bump-AFTER
+    (r2): a full step's combine kernel advances the device seed, so
+    two consecutive steps draw different masks, while a forward
+    WITHOUT a backward reuses the current seed by design."""
     _, net_g, x, tgt = _mk(4, B=256)
     net_g.train()
-    l1 = net_fused_loss(net_g, x.to(DEV), tgt.to(DEV)).item()
-    l2 = net_fused_loss(net_g, x.to(DEV), tgt.to(DEV)).item()
-    assert l1 == l1 and l2 == l2
-    assert l1 != l2   # different dropout draws
     flat = attach_flat_grads(net_g)
-    net_fused_step(net_g, x.to(DEV), tgt.to(DEV))
+    l1 = float(net_fused_step(net_g, x.to(DEV), tgt.to(DEV)))
+    l2 = float(net_fused_step(net_g, x.to(DEV), tgt.to(DEV)))
+    assert l1 == l1 and l2 == l2
+    assert l1 != l2   # the first step's combine advanced the seed
     torch.cuda.synchronize()
     assert flat.abs().sum() > 0
 
