@@ -1349,18 +1349,15 @@ static_assert(T_CONV2 == N_C2K, "conv2 gw tiling: one tile per output "
 template <int NK>
 __device__ __forceinline__ void net_gw_conv2_fold(
     int k0, int tid, int b0, int b1, float* __restrict__ my,
-    const float* __restrict__ p1_ws, const float* __restrict__ ga2_ws,
-    float* __restrict__ sp1_lds,    // 2*N_P1 floats, 16 B aligned
-    float* __restrict__ sg2_lds,    // 2*128 floats, 16 B aligned
-    float* __restrict__ wacc_lds) { // 2*251 floats
-  // LDS is provided by the caller so the NK=1 and NK=2 instantiations
-  // share ONE static allocation (in-template statics doubled the
-  // kernel's LDS and cost occupancy — r2 ledger)
-  float* sp1[2] = {sp1_lds, sp1_lds + N_P1};
-  float* sg2[2] = {sg2_lds, sg2_lds + 64 * NK};
-  float* wacc[NK];
-  #pragma unroll
-  for (int kk = 0; kk < NK; ++kk) wacc[kk] = wacc_lds + kk * 251;
+    const float* __restrict__ p1_ws, const float* __restrict__ ga2_ws) {
+  // statics INSIDE the template: the two instantiations each get an
+  // allocation (the kernel carries both), but a caller-hoisted shared
+  // buffer passed as pointers measured 15% SLOWER end to end (730 vs
+  // 631 us gw-all at B=4096, same box) — the pointer indirection costs
+  // more than the duplicated LDS at this occupancy (r2 A/B, ledger).
+  __shared__ __attribute__((aligned(16))) float sp1[2][N_P1];
+  __shared__ __attribute__((aligned(16))) float sg2[2][64 * NK];
+  __shared__ float wacc[NK][251];
   const int ohp = tid / 50;          // 0..3 (tid < 200)
   const int cr = tid % 50;
   const int c = cr / 5, r = cr % 5;
@@ -1477,16 +1474,11 @@ __device__ __forceinline__ void net_gw_tile(
     //  * single mode (large B): one channel per block — smaller
     //    blocks pack the 2600-block grid without straggler tails,
     //    which beats the re-read saving end to end.
-    __shared__ __attribute__((aligned(16))) float sp1_lds[2 * N_P1];
-    __shared__ __attribute__((aligned(16))) float sg2_lds[2 * 128];
-    __shared__ float wacc_lds[2 * 251];
     if (b1 - b0 <= 16) {
       if (tile < N_C2K / 2)
-        net_gw_conv2_fold<2>(tile * 2, tid, b0, b1, my, p1_ws, ga2_ws,
-                             sp1_lds, sg2_lds, wacc_lds);
+        net_gw_conv2_fold<2>(tile * 2, tid, b0, b1, my, p1_ws, ga2_ws);
     } else {
-      net_gw_conv2_fold<1>(tile, tid, b0, b1, my, p1_ws, ga2_ws,
-                           sp1_lds, sg2_lds, wacc_lds);
+      net_gw_conv2_fold<1>(tile, tid, b0, b1, my, p1_ws, ga2_ws);
     }
     return;
   }
