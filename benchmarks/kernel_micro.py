@@ -22,7 +22,7 @@ from dist_tuto_pth_amd.ops.fused import _ws, attach_flat_grads  # noqa: E402
 from dist_tuto_pth_amd.utils.native import load_native  # noqa: E402
 
 # tile segment table — keep in sync with csrc/kernels.hip
-T_CONV2, T_FC1, T_CONV1, T_FC2 = 20, 63, 8, 2
+T_CONV2, T_FC1, T_CONV1, T_FC2 = 20, 63, 24, 2
 
 
 def time_fn(fn, reps=200, warmup=20):
@@ -92,6 +92,7 @@ def main():
 
     bchunk = (B + 31) // 32
     nch = (B + bchunk - 1) // bchunk
+    c1_ext = (8 if bchunk <= 16 else 24) - 1  # keep in sync: gw_c1_subs
 
     def seg(base, ntiles):
         def f():
@@ -117,14 +118,14 @@ def main():
     results["gw all tiles"] = time_fn(seg(0, T_CONV2 + T_FC1 + T_CONV1 + T_FC2))
     results["gw combine"] = time_fn(lambda: k.net_gw_combine_raw(
         ws["part"].data_ptr(), [p.grad.data_ptr() for p in params], nch,
-        s))
+        c1_ext, s))
     ws.setdefault("loss_part", torch.empty(4096, device=dev))
     bufs = [b.data_ptr() for b in opt._bufs]
     results["gw combine+sgd+loss"] = time_fn(
         lambda: k.net_gw_combine_sgd_raw(
             ws["part"].data_ptr(), [p.grad.data_ptr() for p in params],
-            pp, bufs, nch, 0.01, 0.5, ws["loss_part"].data_ptr(),
-            ws["loss"].data_ptr(), 128, s))
+            pp, bufs, nch, c1_ext, 0.01, 0.5,
+            ws["loss_part"].data_ptr(), ws["loss"].data_ptr(), 128, s))
     results["sgd"] = time_fn(opt.step)
 
     for name, us in results.items():

@@ -1332,13 +1332,13 @@ net_fused_fwdbwd_kernel(
 static_assert(T_CONV2 == N_C2K, "conv2 gw tiling: one tile per output "
               "channel (single mode) or per channel pair (pair mode)");
 #define T_FC1 63          // ceil(16050/256)
-#define T_CONV1 8         // 8 sub-blocks over 3-output-row bands
+#define T_CONV1 24        // one sub-block per conv1 output row
 #define T_FC2 2           // ceil(510/256)
 #define GW_TILES (T_CONV2 + T_FC1 + T_CONV1 + T_FC2)
-// conv1's 8 sub-blocks write disjoint 260-float slices: sub 0 to the
-// canonical [OFF_W1,OFF_B1] region, subs 1-7 to an extension past
+// conv1's 24 sub-blocks write disjoint 260-float slices: sub 0 to the
+// canonical [OFF_W1,OFF_B1] region, subs 1-23 to an extension past
 // GW_TOTAL; the combine kernel folds the extension back in.
-#define GW_ROW (GW_TOTAL + 7 * 260)
+#define GW_ROW (GW_TOTAL + 23 * 260)
 
 // conv2 weight-gradient fold, templated on channels-per-block so every
 // accumulator index is compile-time (a runtime nk spilled q[][] to
@@ -1456,7 +1456,8 @@ __device__ __forceinline__ void net_gw_conv2_fold(
 // laid out like the flat grad buffer.  Shared by net_gw_partial_kernel
 // and the single-launch net_step_kernel.
 __device__ __forceinline__ void net_gw_tile(
-    int tile, int tid, int b0, int b1, float* __restrict__ my,
+    int tile, int tid, int b0, int b1, int c1_subs,
+    float* __restrict__ my,
     const float* __restrict__ x,
     const float* __restrict__ p1_ws,
     const float* __restrict__ p2_ws,
@@ -1486,25 +1487,28 @@ __device__ __forceinline__ void net_gw_tile(
   if (tile < T_CONV1) {  // conv1: gw [10][1][5][5] + gb [10]
     // 250 outputs is too little parallelism for element-per-thread at
     // this cost (24x24 window x batch): split each output over its 24
-    // output rows -> 6000 independent items, LDS-atomic reduce, and
-    // over 8 sub-blocks (3 output rows each) so the grid column's
-    // straggler block shrinks 8x.  Sub 0 writes the canonical region,
-    // subs 1-7 the extension rows summed by the combine kernel.
-    // (r2 note: an LDS-staged double-buffered variant of this tile was
-    // MEASURED SLOWER - 8.7 -> 11.3 us at B=128, 281 -> 325 us at
-    // B=4096: the ~25x grow-row re-reads are L1 hits and the staging
-    // address math + extra barrier cost more than they save.  This
-    // direct-from-L1 form stands; the conv2 tile, whose re-reads miss
-    // L1, DID win from the same treatment.)
+    // output rows, LDS-atomic reduce.  The sub-block count is
+    // ADAPTIVE (r2): 24 single-row subs at large B (this family was
+    // the gw straggler: 3x the blocks, 281 -> 254 us at B=4096) but 8
+    // three-row bands at small B, where the extra 16 extension rows
+    // made the COMBINE 3x slower (8 -> 24 us) for nothing.  Sub 0
+    // writes the canonical region, subs 1..c1_subs-1 the extension
+    // rows; the combine folds c1_subs-1 rows.
+    // (r2 note: an LDS-staged double-buffered variant was MEASURED
+    // SLOWER — the grow/x re-reads are L1 hits and the staging math
+    // cost more than it saved.)
     const int sub = tile;
-    const int oh0 = sub * 3;
+    if (sub >= c1_subs) return;
+    const int rows = 24 / c1_subs;     // 3 (band mode) or 1 (row mode)
+    const int oh0 = sub * rows;
+    const int n_w = 250 * rows;
     __shared__ float wacc[260];
     for (int i = tid; i < 260; i += 256) wacc[i] = 0.f;
     __syncthreads();
-    for (int it = tid; it < 750 + 30; it += 256) {
+    for (int it = tid; it < n_w + 10 * rows; it += 256) {
       float a = 0.f;
-      if (it < 750) {
-        const int e = it / 3, oh = oh0 + it % 3;
+      if (it < n_w) {
+        const int e = it / rows, oh = oh0 + it % rows;
         const int k = e / 25, r = (e / 5) % 5, sx = e % 5;
         for (int b = b0; b < b1; ++b) {
           const float* grow = ga1_ws + (int64_t)b * N_A1 + k * 576 +
@@ -1520,8 +1524,8 @@ __device__ __forceinline__ void net_gw_tile(
         }
         atomicAdd(&wacc[e], a);
       } else {
-        const int j = it - 750;
-        const int k = j / 3, oh = oh0 + j % 3;
+        const int j = it - n_w;
+        const int k = j / rows, oh = oh0 + j % rows;
         for (int b = b0; b < b1; ++b) {
           const float* grow = ga1_ws + (int64_t)b * N_A1 + k * 576 +
                               oh * 24;
@@ -1587,7 +1591,7 @@ __device__ __forceinline__ void net_gw_tile(
 }
 
 __global__ void __launch_bounds__(256)
-net_gw_partial_kernel(const float* __restrict__ x,
+net_gw_partial_kernel(int c1_subs, const float* __restrict__ x,
                       const float* __restrict__ p1_ws,
                       const float* __restrict__ p2_ws,
                       const float* __restrict__ d3_ws,
@@ -1598,7 +1602,7 @@ net_gw_partial_kernel(const float* __restrict__ x,
                       float* __restrict__ part,  // [nch][GW_ROW]
                       int B, int bchunk, int tile_base) {
   const int b0 = blockIdx.y * bchunk;
-  net_gw_tile(blockIdx.x + tile_base, threadIdx.x, b0,
+  net_gw_tile(blockIdx.x + tile_base, threadIdx.x, b0, c1_subs,
               min(B, b0 + bchunk), part + (int64_t)blockIdx.y * GW_ROW,
               x, p1_ws, p2_ws, d3_ws, ga1_ws, ga2_ws, gh1_ws, glog_ws);
 }
@@ -1608,7 +1612,7 @@ net_gw_partial_kernel(const float* __restrict__ x,
 struct GwPtrs { float* p[8]; };
 // sum flat-grad element i over the nch chunk rows (+ conv1 extension)
 __device__ __forceinline__ float net_gw_combine_elem(
-    int i, int nch, const float* __restrict__ part) {
+    int i, int nch, int c1_ext, const float* __restrict__ part) {
   // four independent accumulator chains: the single 32-deep
   // load+add chain was latency-bound (VALUBusy ~0, profiles/)
   float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
@@ -1623,10 +1627,12 @@ __device__ __forceinline__ float net_gw_combine_elem(
   if (i < 260) {  // conv1 sub-block extension rows (see GW_ROW)
     for (int c2 = 0; c2 < nch; ++c2) {
       const float* ext = part + (int64_t)c2 * GW_ROW + GW_TOTAL;
-      a0 += ext[i] + ext[4 * 260 + i];
-      a1 += ext[260 + i] + ext[5 * 260 + i];
-      a2 += ext[2 * 260 + i] + ext[6 * 260 + i];
-      a3 += ext[3 * 260 + i];
+      for (int s = 0; s < c1_ext; s += 4) {
+        a0 += ext[s * 260 + i];
+        if (s + 1 < c1_ext) a1 += ext[(s + 1) * 260 + i];
+        if (s + 2 < c1_ext) a2 += ext[(s + 2) * 260 + i];
+        if (s + 3 < c1_ext) a3 += ext[(s + 3) * 260 + i];
+      }
     }
   }
   return (a0 + a1) + (a2 + a3);
@@ -1639,17 +1645,17 @@ __device__ __forceinline__ float net_gw_combine_elem(
 // ran a latency-exposed 8-deep L2 chain (profiles/).  Returns the full
 // sum on EVERY lane of the quad (butterfly reduction).
 __device__ __forceinline__ float net_gw_combine_elem_quad(
-    int i, int q, int nch, const float* __restrict__ part) {
+    int i, int q, int nch, int c1_ext, const float* __restrict__ part) {
   float a0 = 0.f, a1 = 0.f;
   for (int c = q; c < nch; c += 8)
     a0 += part[(int64_t)c * GW_ROW + i];
   for (int c = q + 4; c < nch; c += 8)
     a1 += part[(int64_t)c * GW_ROW + i];
-  if (i < 260 && q == 0) {  // conv1 extension rows, lane 0 of the quad
+  if (i < 260) {  // conv1 extension rows, spread across the quad
     for (int c2 = 0; c2 < nch; ++c2) {
       const float* ext = part + (int64_t)c2 * GW_ROW + GW_TOTAL;
-      a0 += ext[i] + ext[260 + i] + ext[2 * 260 + i] + ext[3 * 260 + i];
-      a1 += ext[4 * 260 + i] + ext[5 * 260 + i] + ext[6 * 260 + i];
+      for (int s = q; s < c1_ext; s += 8) a0 += ext[s * 260 + i];
+      for (int s = q + 4; s < c1_ext; s += 8) a1 += ext[s * 260 + i];
     }
   }
   float acc = a0 + a1;
@@ -1691,7 +1697,7 @@ __device__ __forceinline__ void net_loss_finalize(
 }
 
 __global__ void net_gw_combine_kernel(const float* __restrict__ part,
-                                      GwPtrs g, int nch,
+                                      GwPtrs g, int nch, int c1_ext,
                                       const float* __restrict__ loss_part,
                                       float* __restrict__ loss_out,
                                       int nblk_fwd,
@@ -1702,7 +1708,7 @@ __global__ void net_gw_combine_kernel(const float* __restrict__ part,
        t4 < (int64_t)GW_TOTAL * 4;
        t4 += (int64_t)gridDim.x * blockDim.x) {
     const int i = (int)(t4 >> 2), q = (int)(t4 & 3);
-    const float acc = net_gw_combine_elem_quad(i, q, nch, part);
+    const float acc = net_gw_combine_elem_quad(i, q, nch, c1_ext, part);
     if (q == 0) {
       const int t = net_gw_tensor_of(i, off);
       g.p[t][i - off[t]] = acc;
@@ -1719,7 +1725,8 @@ __global__ void net_gw_combine_kernel(const float* __restrict__ part,
 // the momentum buffer and the parameter exactly like sgd_step_kernel.
 __global__ void net_gw_combine_sgd_kernel(const float* __restrict__ part,
                                           GwPtrs g, GwPtrs prm,
-                                          GwPtrs buf, int nch, float lr,
+                                          GwPtrs buf, int nch, int c1_ext,
+                                          float lr,
                                           float mu,
                                           const float* __restrict__ loss_part,
                                           float* __restrict__ loss_out,
@@ -1731,7 +1738,7 @@ __global__ void net_gw_combine_sgd_kernel(const float* __restrict__ part,
        t4 < (int64_t)GW_TOTAL * 4;
        t4 += (int64_t)gridDim.x * blockDim.x) {
     const int i = (int)(t4 >> 2), q = (int)(t4 & 3);
-    const float acc = net_gw_combine_elem_quad(i, q, nch, part);
+    const float acc = net_gw_combine_elem_quad(i, q, nch, c1_ext, part);
     if (q == 0) {
       const int t = net_gw_tensor_of(i, off);
       const int64_t j = i - off[t];
@@ -1844,7 +1851,7 @@ net_step_kernel(
   for (int t = wg; t < GW_TILES * nch; t += nblk) {
     const int tile = t % GW_TILES, ch = t / GW_TILES;
     const int b0 = ch * bchunk;
-    net_gw_tile(tile, tid, b0, min(B, b0 + bchunk),
+    net_gw_tile(tile, tid, b0, min(B, b0 + bchunk), 8,
                 part + (int64_t)ch * GW_ROW, x, p1_ws, p2_ws, d3_ws,
                 ga1_ws, ga2_ws, gh1_ws, glog_ws);
     __syncthreads();
@@ -1855,7 +1862,7 @@ net_step_kernel(
   const int off[9] = {OFF_W1, OFF_B1, OFF_W2, OFF_B2, OFF_WF1, OFF_BF1,
                       OFF_WF2, OFF_BF2, GW_TOTAL};
   for (int i = wg * 256 + tid; i < GW_TOTAL; i += nblk * 256) {
-    const float acc = net_gw_combine_elem(i, nch, part);
+    const float acc = net_gw_combine_elem(i, nch, 7, part);
     const int t = net_gw_tensor_of(i, off);
     const int64_t j = i - off[t];
     grd.p[t][j] = acc;
@@ -2127,6 +2134,11 @@ void sgd_step(const std::vector<uintptr_t>& ps,
 }
 
 
+// conv1 sub-block count: 8 three-row bands at small chunks (a cheap
+// combine fold), 24 single-row subs at large chunks (3x the blocks
+// where conv1 is the gw straggler) — r2 ledger.
+static inline int gw_c1_subs(int bchunk) { return bchunk <= 16 ? 8 : 24; }
+
 // gw chunk count: default 32 (or B when smaller); DTP_GW_NCH overrides.
 static int gw_nch(int B) {
   static int env_nch = -2;
@@ -2269,7 +2281,8 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
   const int nch = gw_nch(B);
   const int bchunk = (B + nch - 1) / nch;
   hipLaunchKernelGGL(net_gw_partial_kernel, dim3(GW_TILES, nch),
-                     dim3(256), 0, S(stream), (const float*)x,
+                     dim3(256), 0, S(stream), gw_c1_subs(bchunk),
+                     (const float*)x,
                      (const float*)p1_ws, (const float*)p2_ws,
                      (const float*)d3_ws, (const float*)ga1_ws,
                      (const float*)ga2_ws, (const float*)gh1_ws,
@@ -2283,6 +2296,7 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
   hipLaunchKernelGGL(net_gw_combine_kernel,
                      dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
                      S(stream), (const float*)part_ws, gp, nch,
+                     gw_c1_subs(bchunk) - 1,
                      (const float*)loss_part, (float*)loss_out,
                      grid_for(B, 1),
                      (training && seed_dev)
@@ -2326,7 +2340,8 @@ void net_fused_fwdbwd(
   const int nch = gw_nch(B);
   const int bchunk = (B + nch - 1) / nch;
   hipLaunchKernelGGL(net_gw_partial_kernel, dim3(GW_TILES, nch),
-                     dim3(256), 0, S(stream), (const float*)x,
+                     dim3(256), 0, S(stream), gw_c1_subs(bchunk),
+                     (const float*)x,
                      (const float*)p1_ws, (const float*)p2_ws,
                      (const float*)d3_ws, (const float*)ga1_ws,
                      (const float*)ga2_ws, (const float*)gh1_ws,
@@ -2340,6 +2355,7 @@ void net_fused_fwdbwd(
     hipLaunchKernelGGL(net_gw_combine_kernel,
                        dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
                        S(stream), (const float*)part_ws, gp, nch,
+                       gw_c1_subs(bchunk) - 1,
                        (const float*)loss_part, (float*)loss_out,
                        nblk, sb);
   } else {
@@ -2351,6 +2367,7 @@ void net_fused_fwdbwd(
     hipLaunchKernelGGL(net_gw_combine_sgd_kernel,
                        dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
                        S(stream), (const float*)part_ws, gp, pp, bp, nch,
+                       gw_c1_subs(bchunk) - 1,
                        (float)lr, (float)mu, (const float*)loss_part,
                        (float*)loss_out, nblk, sb);
   }
@@ -2395,7 +2412,8 @@ void net_fused_bwd_sgd(uintptr_t x, uintptr_t w2, uintptr_t wf1,
   const int nch = gw_nch(B);
   const int bchunk = (B + nch - 1) / nch;
   hipLaunchKernelGGL(net_gw_partial_kernel, dim3(GW_TILES, nch),
-                     dim3(256), 0, S(stream), (const float*)x,
+                     dim3(256), 0, S(stream), gw_c1_subs(bchunk),
+                     (const float*)x,
                      (const float*)p1_ws, (const float*)p2_ws,
                      (const float*)d3_ws, (const float*)ga1_ws,
                      (const float*)ga2_ws, (const float*)gh1_ws,
@@ -2410,6 +2428,7 @@ void net_fused_bwd_sgd(uintptr_t x, uintptr_t w2, uintptr_t wf1,
   hipLaunchKernelGGL(net_gw_combine_sgd_kernel,
                      dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
                      S(stream), (const float*)part_ws, gp, pp, bp, nch,
+                     gw_c1_subs(bchunk) - 1,
                      (float)lr, (float)mu, (const float*)loss_part,
                      (float*)loss_out, grid_for(B, 1),
                      (training && seed_dev)
@@ -2420,13 +2439,13 @@ void net_fused_bwd_sgd(uintptr_t x, uintptr_t w2, uintptr_t wf1,
 // in isolation)
 void net_gw_combine_raw(uintptr_t part_ws,
                         const std::vector<uintptr_t>& grd_v, int nch,
-                        uintptr_t stream) {
+                        int c1_ext, uintptr_t stream) {
   GwPtrs gp{};
   for (int i = 0; i < 8; ++i) gp.p[i] = (float*)grd_v[i];
   hipLaunchKernelGGL(net_gw_combine_kernel,
                      dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
-                     S(stream), (const float*)part_ws, gp, nch, nullptr,
-                     nullptr, 0, nullptr);
+                     S(stream), (const float*)part_ws, gp, nch, c1_ext,
+                     nullptr, nullptr, 0, nullptr);
 }
 
 // raw combine+sgd launch (microbenchmarks)
@@ -2434,7 +2453,7 @@ void net_gw_combine_sgd_raw(uintptr_t part_ws,
                             const std::vector<uintptr_t>& grd_v,
                             const std::vector<uintptr_t>& prm_v,
                             const std::vector<uintptr_t>& buf_v,
-                            int nch, double lr, double mu,
+                            int nch, int c1_ext, double lr, double mu,
                             uintptr_t loss_part, uintptr_t loss_out,
                             int nblk_fwd, uintptr_t stream) {
   GwPtrs gp{}, pp{}, bp{};
@@ -2446,8 +2465,9 @@ void net_gw_combine_sgd_raw(uintptr_t part_ws,
   hipLaunchKernelGGL(net_gw_combine_sgd_kernel,
                      dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
                      S(stream), (const float*)part_ws, gp, pp, bp, nch,
-                     (float)lr, (float)mu, (const float*)loss_part,
-                     (float*)loss_out, nblk_fwd, nullptr);
+                     c1_ext, (float)lr, (float)mu,
+                     (const float*)loss_part, (float*)loss_out,
+                     nblk_fwd, nullptr);
 }
 
 // raw tile-segment launch of the partial weight-gradient kernel
@@ -2459,11 +2479,12 @@ void net_gw_partial_raw(uintptr_t x, uintptr_t p1_ws, uintptr_t p2_ws,
                         int bchunk, int tile_base, int ntiles, int nch,
                         uintptr_t stream) {
   hipLaunchKernelGGL(net_gw_partial_kernel, dim3(ntiles, nch), dim3(256),
-                     0, S(stream), (const float*)x, (const float*)p1_ws,
-                     (const float*)p2_ws, (const float*)d3_ws,
-                     (const float*)ga1_ws, (const float*)ga2_ws,
-                     (const float*)gh1_ws, (const float*)glog_ws,
-                     (float*)part_ws, B, bchunk, tile_base);
+                     0, S(stream), gw_c1_subs(bchunk), (const float*)x,
+                     (const float*)p1_ws, (const float*)p2_ws,
+                     (const float*)d3_ws, (const float*)ga1_ws,
+                     (const float*)ga2_ws, (const float*)gh1_ws,
+                     (const float*)glog_ws, (float*)part_ws, B, bchunk,
+                     tile_base);
 }
 
 // cooperative grid-barrier cost probe: `nsync` grid.sync()s and

@@ -5,7 +5,7 @@ one kernel for the entire forward (conv1..log_softmax+NLL,
 train_dist.py:64-71 + :120, conv+pool register-fused, per-block loss
 partials), one for the data backward (sibling-workgroup split,
 4-wide register-blocked transposed conv), one segmented partial
-weight-gradient kernel ([conv2|fc1|conv1x8|fc2] tiles x batch chunks),
+weight-gradient kernel ([conv2|conv1x24|fc1|fc2] tiles x batch chunks),
 and one combine kernel that also finalizes the loss, advances the
 dropout seed, and (single-GPU) applies the SGD+momentum update.
 rocprof evidence and the optimization ladder live in profiles/.
@@ -45,9 +45,10 @@ def _ws(B: int, device) -> Dict[str, torch.Tensor]:
             "p2": f(B, 320), "idx2": u8(B, 320), "h1": f(B, 50),
             "m3": u8(B, 50), "d3": f(B, 50), "logp": f(B, 10),
             "glog": f(B, 10), "gh1": f(B, 50), "ga2": f(B, 1280),
-            # partial rows are GW_ROW = 21840 + 7*260 wide (conv1's 8
-            # weight-grad sub-blocks write disjoint slices; kernels.hip)
-            "ga1": f(B, 5760), "part": f(32, 23660), "loss": f(()),
+            # partial rows are GW_ROW = 21840 + 23*260 wide (conv1's
+            # 24 weight-grad sub-blocks write disjoint slices;
+            # kernels.hip)
+            "ga1": f(B, 5760), "part": f(32, 27820), "loss": f(()),
             # loss_part must cover the LARGEST forward grid any caller
             # can launch (grid_for(B,1) <= 4096) — sized once here so
             # every entry point shares a safe buffer regardless of
