@@ -1602,8 +1602,9 @@ net_gw_partial_kernel(int c1_subs, const float* __restrict__ x,
                       float* __restrict__ part,  // [nch][GW_ROW]
                       int B, int bchunk, int tile_base) {
   const int b0 = blockIdx.y * bchunk;
-  net_gw_tile(blockIdx.x + tile_base, threadIdx.x, b0, c1_subs,
-              min(B, b0 + bchunk), part + (int64_t)blockIdx.y * GW_ROW,
+  net_gw_tile(blockIdx.x + tile_base, threadIdx.x, b0,
+              min(B, b0 + bchunk), c1_subs,
+              part + (int64_t)blockIdx.y * GW_ROW,
               x, p1_ws, p2_ws, d3_ws, ga1_ws, ga2_ws, gh1_ws, glog_ws);
 }
 
