@@ -4,9 +4,14 @@
 
 PY ?= python3
 
-.PHONY: all build test test-gpu ptp sendrecv allreduce train bench clean
+.PHONY: all build docs test test-gpu ptp sendrecv allreduce train bench clean
 
-all: build
+all: build docs
+
+# render the tutorial to HTML (the reference's `all` target renders
+# tuto.md -> tuto.html + index.html via its paperify.py)
+docs:
+	$(PY) docs/paperify.py
 
 build:
 	$(PY) build.py

@@ -48,3 +48,22 @@ def test_train_dist_example():
                          timeout=240)
     assert out.returncode == 0, out.stderr[-2000:]
     assert "epoch 0" in out.stdout
+
+
+def test_paperify_renders_tutorial(tmp_path):
+    """The docs build (the reference Makefile's `all` target): render
+    TUTORIAL.md to tutorial.html + byte-identical index.html."""
+    import subprocess
+    import sys as _sys
+    import os as _os
+    root = _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))
+    r = subprocess.run([_sys.executable, "docs/paperify.py"], cwd=root,
+                       capture_output=True, text=True, timeout=60)
+    assert r.returncode == 0, r.stderr
+    with open(_os.path.join(root, "docs", "tutorial.html")) as f:
+        h = f.read()
+    for sec in ("Communication backends", "Initialization methods",
+                "<table>", "<pre><code>"):
+        assert sec in h, sec
+    with open(_os.path.join(root, "docs", "index.html")) as f:
+        assert f.read() == h
