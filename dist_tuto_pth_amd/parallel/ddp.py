@@ -67,7 +67,14 @@ class _Bucket:
         for p, off, n in zip(self.params, self.offsets, self.numels):
             sl = self.flat[off:off + n]
             if p.grad is None or p.grad.data_ptr() != sl.data_ptr():
-                g = sl.view_as(p)
+                if p.is_contiguous():
+                    g = sl.view_as(p)
+                else:
+                    # match the parameter's memory format (e.g.
+                    # channels_last) so autograd accumulates straight
+                    # into the bucket instead of reducing through a
+                    # layout-contract copy each step
+                    g = sl.as_strided(p.shape, p.stride())
                 if p.grad is not None:
                     g.copy_(p.grad.reshape(-1).view_as(p))
                 else:

@@ -118,3 +118,21 @@ def test_ddp_zero_copy_views():
 
 def test_average_gradients_math():
     launch(_fn_avg_matches_manual, 2, timeout=120)
+
+
+def test_bucket_views_match_param_memory_format():
+    """channels_last params get stride-matched bucket views so autograd
+    accumulates straight into the bucket (no layout-contract copy)."""
+    import torch as th
+    from dist_tuto_pth_amd.parallel.ddp import _Bucket
+    m = th.nn.Conv2d(4, 8, 3).to(memory_format=th.channels_last)
+    b = _Bucket(list(m.parameters()), th.float32, th.device("cpu"))
+    x = th.randn(2, 4, 8, 8).to(memory_format=th.channels_last)
+    m(x).sum().backward()
+    w = m.weight
+    assert w.grad.data_ptr() == b.flat[b.offsets[0]:].data_ptr()
+    assert w.grad.stride() == w.stride()
+    m2 = th.nn.Conv2d(4, 8, 3)
+    m2.load_state_dict(m.state_dict())
+    m2(x.contiguous()).sum().backward()
+    assert th.allclose(w.grad, m2.weight.grad, atol=1e-5)
