@@ -57,11 +57,28 @@ def _check(outdir, prefix, size):
         assert abs(a - b) < 0.2
 
 
+def _fn_train_tcp(rank, size):
+    ds = SyntheticMNIST(n=512, seed=1234)
+    losses, model = training.run(rank, size, epochs=2, device="cpu",
+                                 dataset=ds, batch_size=128)
+    _save(rank, losses, model, "t")
+
+
 def test_sync_sgd_average_gradients_world2():
     with tempfile.TemporaryDirectory() as d:
         os.environ["_TRAIN_OUT"] = d
         launch(_fn_train, 2, timeout=300)
         _check(d, "r", 2)
+
+
+def test_sync_sgd_on_native_tcp_backend():
+    """The full training loop over this repo's OWN wire (tcp socket
+    mesh + p2p-composed collectives): replicas stay identical, losses
+    decrease — torch.distributed is never involved."""
+    with tempfile.TemporaryDirectory() as d:
+        os.environ["_TRAIN_OUT"] = d
+        launch(_fn_train_tcp, 2, backend="tcp", timeout=300)
+        _check(d, "t", 2)
 
 
 def test_sync_sgd_ddp_world2():
