@@ -146,16 +146,38 @@ class _RcclBackend:
     def _seq_key(self, dst, src, seq):
         return f"p2p:n:{self._ptag}:{dst}:{src}:{seq}"
 
+    def _notify_send(self, dst):
+        """Post the any-source sequence notification WITHOUT putting
+        the store round-trip on the send's critical path: a single
+        daemon thread drains a queue, which preserves per-pair
+        sequence order (the only ordering the protocol needs)."""
+        import queue as _q
+        import threading as _t
+        seqs = getattr(self, "_send_seq", None)
+        if seqs is None:
+            seqs = self._send_seq = {}
+            self._notify_q = _q.Queue()
+
+            def _drain():
+                while True:
+                    key = self._notify_q.get()
+                    if key is None:
+                        return
+                    try:
+                        self._store.add(key, 1)
+                    except Exception:
+                        return
+            self._notify_t = _t.Thread(target=_drain, daemon=True)
+            self._notify_t.start()
+        s = seqs.get(dst, 0)
+        seqs[dst] = s + 1
+        self._notify_q.put(self._seq_key(dst, self._rank, s))
+
     def send(self, t, dst, blocking):
         from . import Work
         _check(t)
         if self._store is not None:
-            seqs = getattr(self, "_send_seq", None)
-            if seqs is None:
-                seqs = self._send_seq = {}
-            s = seqs.get(dst, 0)
-            seqs[dst] = s + 1
-            self._store.add(self._seq_key(dst, self._rank, s), 1)
+            self._notify_send(dst)
         self._comm.send(t.data_ptr(), t.numel(), _DTYPE[t.dtype], dst,
                         _stream())
         if blocking:
