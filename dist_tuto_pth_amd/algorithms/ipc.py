@@ -56,7 +56,7 @@ class IpcTransport:
         self.tag = tag
         self._phase = 0
         nrows = max(size - 1, 1)
-        self.buf = rx.DeviceBuffer(nrows * row_bytes, device)
+        self.buf = rx.DeviceBuffer(nrows * self.row_bytes, device)
         store.set(f"{tag}:h:{rank}", self.buf.ipc_handle())
         self.peer_base: Dict[int, int] = {}
         for p in range(size):
