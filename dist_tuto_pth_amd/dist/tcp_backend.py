@@ -83,6 +83,14 @@ class _TcpBackend:
         self._world = world_size
 
     def destroy(self):
+        # drain in lockstep so no rank closes its pair sockets while a
+        # peer is still mid-collective (the gloo delegation does the
+        # same with its barrier)
+        if self._mesh is not None and self._world > 1:
+            try:
+                self.barrier()
+            except Exception:
+                pass
         self._mesh = None
 
     # ------------------------------------------------------------------
@@ -125,11 +133,14 @@ class _TcpBackend:
         self.recv(recv_t, src, blocking=True)
         req.wait()
 
-    # internal raw helpers
+    # internal raw helpers (zero-copy: tensors must be contiguous so
+    # the mesh reads/writes their memory directly)
     def _send_raw(self, t, peer):
+        _check_cpu(t)
         self._mesh.send(peer, t.data_ptr(), _nbytes(t))
 
     def _recv_raw(self, t, peer):
+        _check_cpu(t)
         self._mesh.recv(peer, t.data_ptr(), _nbytes(t))
 
     # ------------------------------------------------------------------
