@@ -1136,7 +1136,7 @@ __device__ __forceinline__ void net_bwd_sample(
       }
     }
     __syncthreads();
-    for (int i = tid; i < N_C2K * 256; i += 256) gd2p[i] = 0.f;
+    for (int i = tid; i < N_C2K * 272; i += 256) gd2p[i] = 0.f;
     __syncthreads();
     for (int i = tid; i < N_A2; i += 256) {
       const int k = i / 64;
@@ -1146,7 +1146,7 @@ __device__ __forceinline__ void net_bwd_sample(
       if (half == 0)
         ga2_ws[(int64_t)b * N_A2 + i] = g;
       const int oh = (i / 8) % 8, ow = i % 8;
-      gd2p[k * 256 + (oh + 4) * 16 + (ow + 4)] = g;
+      gd2p[k * 272 + (oh + 4) * 17 + (ow + 4)] = g;
     }
     __syncthreads();
 
@@ -1166,11 +1166,11 @@ __device__ __forceinline__ void net_bwd_sample(
       const int h = rem / 3, wc0 = (rem % 3) * 4;
       float q0 = 0.f, q1 = 0.f, q2 = 0.f, q3 = 0.f;
       for (int k = 0; k < N_C2K; ++k) {
-        const float* gk = gd2p + k * 256;
+        const float* gk = gd2p + k * 272;
         const float* wk = w2s + (k * 10 + c) * 25;
         #pragma unroll
         for (int r = 0; r < 5; ++r) {
-          const float* row = gk + (h - r + 4) * 16 + wc0;
+          const float* row = gk + (h - r + 4) * 17 + wc0;
           const float w0 = wk[r * 5 + 0], w1 = wk[r * 5 + 1];
           const float w2v = wk[r * 5 + 2], w3 = wk[r * 5 + 3];
           const float w4 = wk[r * 5 + 4];
@@ -1230,11 +1230,14 @@ net_fused_bwd_kernel(
   __shared__ float gh1[N_H1];
   __shared__ float gp2[N_P2];
   __shared__ float gd2[N_A2];
-  // zero-padded conv2-out grad [k][16][16]: entry (k,oh+4,ow+4) holds
+  // zero-padded conv2-out grad [k][16][17]: entry (k,oh+4,ow+4) holds
   // gd2[k][oh][ow]; the pad makes the transposed-conv window reads
   // branch-free so the 500-FMA loop unrolls with ILP (the branchy form
-  // was latency-bound at 1 wave/SIMD).
-  __shared__ float gd2p[N_C2K * 256];
+  // was latency-bound at 1 wave/SIMD).  Row stride 17 (not 16): the
+  // power-of-2 stride put same-h-group lanes on the same LDS banks —
+  // PMC measured 59.8% LDSBankConflict on this kernel at B=4096
+  // (profiles/pmc_r2.md); the odd stride de-banks it.
+  __shared__ float gd2p[N_C2K * 272];
   const int tid = threadIdx.x;
 
   for (int i = tid; i < N_C2K * 250; i += 256) w2s[i] = w2[i];
@@ -1278,7 +1281,7 @@ net_fused_fwdbwd_kernel(
     float* __restrict__ loss_part,
     const unsigned long long* __restrict__ seed_p, int B, int training) {
   __shared__ __attribute__((aligned(16))) float w2s[N_C2K * 250 + N_C2K];
-  __shared__ __attribute__((aligned(16))) float pool[6848];
+  __shared__ __attribute__((aligned(16))) float pool[7168];
   const int tid = threadIdx.x;
   const uint64_t seed = seed_p[0];
   // forward carve
@@ -1289,12 +1292,12 @@ net_fused_fwdbwd_kernel(
   float* d3 = pool + 2808;       // 50
   float* logits = pool + 2860;   // 10
   // backward carve (aliased onto the forward's dead buffers)
-  float* gd2p = pool;            // 5120
-  float* gd2 = pool + 5120;      // 1280
-  float* gp2 = pool + 6400;      // 320
-  float* glg = pool + 6720;      // 10
-  float* gd3 = pool + 6732;      // 50
-  float* gh1 = pool + 6784;      // 50 -> 6834 (buffer 6848)
+  float* gd2p = pool;            // 5440 (= 20 k-planes x 16 x 17)
+  float* gd2 = pool + 5440;      // 1280
+  float* gp2 = pool + 6720;      // 320
+  float* glg = pool + 7040;      // 10
+  float* gd3 = pool + 7052;      // 50
+  float* gh1 = pool + 7104;      // 50 -> 7154 (buffer 7168)
   const float sc = 1.f / B;      // dLoss == 1 by construction
 
   float lsum = 0.f;
@@ -1829,7 +1832,7 @@ net_step_kernel(
   // LDS union: bwd carve (11,872 floats) reused by the fwd carve
   // (7,884) and the final-phase reduction scratch; phases are
   // separated by grid barriers.
-  __shared__ __attribute__((aligned(16))) float smem[11872];
+  __shared__ __attribute__((aligned(16))) float smem[12192];
   float* xs = smem;             // 784
   float* w1s = xs + 784;        // 260
   float* p1 = w1s + 260;        // 1440
@@ -1843,7 +1846,7 @@ net_step_kernel(
   float* b_gh1 = smem + 5088;   // 50
   float* b_gp2 = smem + 5152;   // 320
   float* b_gd2 = smem + 5472;   // 1280
-  float* b_gd2p = smem + 6752;  // 5120
+  float* b_gd2p = smem + 6752;  // 5440 (20 x 16 x 17)
   const int tid = threadIdx.x;
   const int wg = blockIdx.x, nblk = gridDim.x;
   cg::grid_group grid = cg::this_grid();
