@@ -1070,9 +1070,9 @@ __device__ __forceinline__ void net_bwd_sample(
     const float* __restrict__ logp_ws,
     float* __restrict__ glog_ws, float* __restrict__ gh1_ws,
     float* __restrict__ ga2_ws, float* __restrict__ ga1_ws,
-    // LDS carve (sizes: 5000, 10, 50, 50, 320, 1280, 5120)
+    // LDS carve (sizes: 5000, 10, 50, 50, 320, 5440)
     const float* w2s, float* glg, float* gd3, float* gh1, float* gp2,
-    float* gd2, float* gd2p) {
+    float* gd2p) {
   {
     // g_logits = (exp(logp) - onehot) * gl / B
     __syncthreads();
@@ -1122,31 +1122,31 @@ __device__ __forceinline__ void net_bwd_sample(
     }
     __syncthreads();
 
-    // pool2 bwd + dropout2d bwd -> g_a2 (conv2 out grad)
-    for (int i = tid; i < N_A2; i += 256) gd2[i] = 0.f;
+    // pool2 bwd + dropout2d bwd -> g_a2, scattered STRAIGHT into the
+    // padded gd2p tile with dropout applied (r2: the old path staged
+    // through a dense gd2 buffer — zero + scatter + full re-read, an
+    // extra 2.5k LDS ops per sample for nothing; gd2 is gone)
+    for (int i = tid; i < N_C2K * 272; i += 256) gd2p[i] = 0.f;
     __syncthreads();
     for (int i = tid; i < N_P2; i += 256) {
       const int c = i / 16, oh = (i / 4) % 4, ow = i % 4;
       const uint8_t v = idx2_ws[(int64_t)b * N_P2 + i];
       if (!(v & 4)) {
         const int am = v & 3;
-        const int base = c * 64 + oh * 2 * 8 + ow * 2;
-        const int off = (am & 1) + (am >> 1) * 8;
-        gd2[base + off] = gp2[i];
+        const int fr = oh * 2 + (am >> 1), fc = ow * 2 + (am & 1);
+        float g = gp2[i];
+        if (training)
+          g = m2_ws[(int64_t)b * N_C2K + c] ? g * 2.f : 0.f;
+        gd2p[c * 272 + (fr + 4) * 17 + (fc + 4)] = g;
       }
     }
     __syncthreads();
-    for (int i = tid; i < N_C2K * 272; i += 256) gd2p[i] = 0.f;
-    __syncthreads();
-    for (int i = tid; i < N_A2; i += 256) {
-      const int k = i / 64;
-      float g = gd2[i];
-      if (training)
-        g = m2_ws[(int64_t)b * N_C2K + k] ? g * 2.f : 0.f;
-      if (half == 0)
-        ga2_ws[(int64_t)b * N_A2 + i] = g;
-      const int oh = (i / 8) % 8, ow = i % 8;
-      gd2p[k * 272 + (oh + 4) * 17 + (ow + 4)] = g;
+    if (half == 0) {  // stash the dense conv2-out grad for the gw pass
+      for (int i = tid; i < N_A2; i += 256) {
+        const int k = i / 64, oh = (i / 8) % 8, ow = i % 8;
+        ga2_ws[(int64_t)b * N_A2 + i] =
+            gd2p[k * 272 + (oh + 4) * 17 + (ow + 4)];
+      }
     }
     __syncthreads();
 
@@ -1229,7 +1229,6 @@ net_fused_bwd_kernel(
   __shared__ float gd3[N_H1];
   __shared__ float gh1[N_H1];
   __shared__ float gp2[N_P2];
-  __shared__ float gd2[N_A2];
   // zero-padded conv2-out grad [k][16][17]: entry (k,oh+4,ow+4) holds
   // gd2[k][oh][ow]; the pad makes the transposed-conv window reads
   // branch-free so the 500-FMA loop unrolls with ILP (the branchy form
@@ -1247,7 +1246,7 @@ net_fused_bwd_kernel(
     net_bwd_sample(bb / split, bb % split, split, tid, B, training, sc,
                    wf1, wf2, tgt, idx1_ws, m2_ws, idx2_ws, h1_ws, m3_ws,
                    logp_ws, glog_ws, gh1_ws, ga2_ws, ga1_ws,
-                   w2s, glg, gd3, gh1, gp2, gd2, gd2p);
+                   w2s, glg, gd3, gh1, gp2, gd2p);
   }
 }
 
@@ -1293,11 +1292,10 @@ net_fused_fwdbwd_kernel(
   float* logits = pool + 2860;   // 10
   // backward carve (aliased onto the forward's dead buffers)
   float* gd2p = pool;            // 5440 (= 20 k-planes x 16 x 17)
-  float* gd2 = pool + 5440;      // 1280
-  float* gp2 = pool + 6720;      // 320
-  float* glg = pool + 7040;      // 10
-  float* gd3 = pool + 7052;      // 50
-  float* gh1 = pool + 7104;      // 50 -> 7154 (buffer 7168)
+  float* gp2 = pool + 5440;      // 320
+  float* glg = pool + 5760;      // 10
+  float* gd3 = pool + 5772;      // 50
+  float* gh1 = pool + 5824;      // 50 -> 5874 (buffer 7168 has slack)
   const float sc = 1.f / B;      // dLoss == 1 by construction
 
   float lsum = 0.f;
@@ -1311,7 +1309,7 @@ net_fused_fwdbwd_kernel(
     net_bwd_sample(b, 0, 1, tid, B, training, sc, wf1, wf2, tgt,
                    idx1_ws, m2_ws, idx2_ws, h1_ws, m3_ws, logp_ws,
                    glog_ws, gh1_ws, ga2_ws, ga1_ws,
-                   w2s, glg, gd3, gh1, gp2, gd2, gd2p);
+                   w2s, glg, gd3, gh1, gp2, gd2p);
     __syncthreads();  // bwd LDS dead before the next sample reuses pool
   }
   if (tid == 0) loss_part[blockIdx.x] = lsum;
@@ -1845,8 +1843,7 @@ net_step_kernel(
   float* b_gd3 = smem + 5024;   // 50
   float* b_gh1 = smem + 5088;   // 50
   float* b_gp2 = smem + 5152;   // 320
-  float* b_gd2 = smem + 5472;   // 1280
-  float* b_gd2p = smem + 6752;  // 5440 (20 x 16 x 17)
+  float* b_gd2p = smem + 5472;  // 5440 (20 x 16 x 17) -> 10912
   const int tid = threadIdx.x;
   const int wg = blockIdx.x, nblk = gridDim.x;
   cg::grid_group grid = cg::this_grid();
@@ -1876,7 +1873,7 @@ net_step_kernel(
     net_bwd_sample(bb / split, bb % split, split, tid, B, training, sc,
                    wf1, wf2, tgt, idx1_ws, m2_ws, idx2_ws, h1_ws, m3_ws,
                    logp_ws, glog_ws, gh1_ws, ga2_ws, ga1_ws,
-                   b_w2s, b_glg, b_gd3, b_gh1, b_gp2, b_gd2, b_gd2p);
+                   b_w2s, b_glg, b_gd3, b_gh1, b_gp2, b_gd2p);
   }
   grid.sync();
 
