@@ -1070,7 +1070,7 @@ __device__ __forceinline__ void net_bwd_sample(
     const float* __restrict__ logp_ws,
     float* __restrict__ glog_ws, float* __restrict__ gh1_ws,
     float* __restrict__ ga2_ws, float* __restrict__ ga1_ws,
-    // LDS carve (sizes: 5000, 10, 50, 50, 320, 5440)
+    // LDS carve (sizes: 5000, 10, 50, 50, 320, 6080)
     const float* w2s, float* glg, float* gd3, float* gh1, float* gp2,
     float* gd2p) {
   {
@@ -1126,7 +1126,7 @@ __device__ __forceinline__ void net_bwd_sample(
     // padded gd2p tile with dropout applied (r2: the old path staged
     // through a dense gd2 buffer — zero + scatter + full re-read, an
     // extra 2.5k LDS ops per sample for nothing; gd2 is gone)
-    for (int i = tid; i < N_C2K * 272; i += 256) gd2p[i] = 0.f;
+    for (int i = tid; i < N_C2K * 304; i += 256) gd2p[i] = 0.f;
     __syncthreads();
     for (int i = tid; i < N_P2; i += 256) {
       const int c = i / 16, oh = (i / 4) % 4, ow = i % 4;
@@ -1137,7 +1137,7 @@ __device__ __forceinline__ void net_bwd_sample(
         float g = gp2[i];
         if (training)
           g = m2_ws[(int64_t)b * N_C2K + c] ? g * 2.f : 0.f;
-        gd2p[c * 272 + (fr + 4) * 17 + (fc + 4)] = g;
+        gd2p[c * 304 + (fr + 4) * 19 + (fc + 4)] = g;
       }
     }
     __syncthreads();
@@ -1145,7 +1145,7 @@ __device__ __forceinline__ void net_bwd_sample(
       for (int i = tid; i < N_A2; i += 256) {
         const int k = i / 64, oh = (i / 8) % 8, ow = i % 8;
         ga2_ws[(int64_t)b * N_A2 + i] =
-            gd2p[k * 272 + (oh + 4) * 17 + (ow + 4)];
+            gd2p[k * 304 + (oh + 4) * 19 + (ow + 4)];
       }
     }
     __syncthreads();
@@ -1166,11 +1166,11 @@ __device__ __forceinline__ void net_bwd_sample(
       const int h = rem / 3, wc0 = (rem % 3) * 4;
       float q0 = 0.f, q1 = 0.f, q2 = 0.f, q3 = 0.f;
       for (int k = 0; k < N_C2K; ++k) {
-        const float* gk = gd2p + k * 272;
+        const float* gk = gd2p + k * 304;
         const float* wk = w2s + (k * 10 + c) * 25;
         #pragma unroll
         for (int r = 0; r < 5; ++r) {
-          const float* row = gk + (h - r + 4) * 17 + wc0;
+          const float* row = gk + (h - r + 4) * 19 + wc0;
           const float w0 = wk[r * 5 + 0], w1 = wk[r * 5 + 1];
           const float w2v = wk[r * 5 + 2], w3 = wk[r * 5 + 3];
           const float w4 = wk[r * 5 + 4];
@@ -1229,14 +1229,14 @@ net_fused_bwd_kernel(
   __shared__ float gd3[N_H1];
   __shared__ float gh1[N_H1];
   __shared__ float gp2[N_P2];
-  // zero-padded conv2-out grad [k][16][17]: entry (k,oh+4,ow+4) holds
-  // gd2[k][oh][ow]; the pad makes the transposed-conv window reads
-  // branch-free so the 500-FMA loop unrolls with ILP (the branchy form
-  // was latency-bound at 1 wave/SIMD).  Row stride 17 (not 16): the
-  // power-of-2 stride put same-h-group lanes on the same LDS banks —
-  // PMC measured 59.8% LDSBankConflict on this kernel at B=4096
-  // (profiles/pmc_r2.md); the odd stride de-banks it.
-  __shared__ float gd2p[N_C2K * 272];
+  // zero-padded conv2-out grad [k][16][19]: entry (k,oh+4,ow+4) holds
+  // the conv2-out grad at [k][oh][ow]; the pad makes the transposed-
+  // conv window reads branch-free so the 500-FMA loop unrolls with
+  // ILP.  Row stride 19 (not 16): PMC measured 59.8% LDSBankConflict
+  // at B=4096 from the power-of-2 stride (profiles/pmc_r2.md); over
+  // the loop's (h, wc0) lane pattern a brute-force scan shows stride
+  // 19 is conflict-FREE on 64 banks (17 still left 3-way clusters).
+  __shared__ float gd2p[N_C2K * 304];
   const int tid = threadIdx.x;
 
   for (int i = tid; i < N_C2K * 250; i += 256) w2s[i] = w2[i];
@@ -1291,11 +1291,11 @@ net_fused_fwdbwd_kernel(
   float* d3 = pool + 2808;       // 50
   float* logits = pool + 2860;   // 10
   // backward carve (aliased onto the forward's dead buffers)
-  float* gd2p = pool;            // 5440 (= 20 k-planes x 16 x 17)
-  float* gp2 = pool + 5440;      // 320
-  float* glg = pool + 5760;      // 10
-  float* gd3 = pool + 5772;      // 50
-  float* gh1 = pool + 5824;      // 50 -> 5874 (buffer 7168 has slack)
+  float* gd2p = pool;            // 6080 (= 20 k-planes x 16 x 19)
+  float* gp2 = pool + 6080;      // 320
+  float* glg = pool + 6400;      // 10
+  float* gd3 = pool + 6412;      // 50
+  float* gh1 = pool + 6464;      // 50 -> 6514 (buffer 7168)
   const float sc = 1.f / B;      // dLoss == 1 by construction
 
   float lsum = 0.f;
@@ -1843,7 +1843,7 @@ net_step_kernel(
   float* b_gd3 = smem + 5024;   // 50
   float* b_gh1 = smem + 5088;   // 50
   float* b_gp2 = smem + 5152;   // 320
-  float* b_gd2p = smem + 5472;  // 5440 (20 x 16 x 17) -> 10912
+  float* b_gd2p = smem + 5472;  // 6080 (20 x 16 x 19) -> 11552
   const int tid = threadIdx.x;
   const int wg = blockIdx.x, nblk = gridDim.x;
   cg::grid_group grid = cg::this_grid();
