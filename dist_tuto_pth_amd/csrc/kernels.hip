@@ -891,68 +891,43 @@ __device__ __forceinline__ float net_fwd_sample(
     }
     __syncthreads();
 
-    // conv2 + dropout2d + pool2 + relu, register-blocked 2-WIDE (r2):
-    // one thread owns TWO horizontally-adjacent pooled cells, so each
-    // (c, r) iteration loads one 8-value strip per input row and
-    // feeds 40 FMA from registers (the one-cell form re-read
-    // overlapping window values from LDS — 2.2x the ds_read count;
-    // PMC showed fwd at VALUBusy 51% / latency-chain-bound, and both
-    // regimes want fewer LDS issues + more independent chains).
-    for (int i = tid; i < N_P2 / 2; i += 256) {
-      const int k = i / 8, ph = (i / 2) % 4, pw0 = (i % 2) * 2;
+    // conv2 + dropout2d + pool2 + relu, same register-blocked shape:
+    // one thread per pooled cell, four conv outputs in four chains,
+    // channel dropout applied in registers before the max.
+    for (int i = tid; i < N_P2; i += 256) {
+      const int k = i / 16, ph = (i / 4) % 4, pw = i % 4;
       const float bias = w2s[N_C2K * 250 + k];
-      float qa00 = bias, qa01 = bias, qa10 = bias, qa11 = bias;
-      float qb00 = bias, qb01 = bias, qb10 = bias, qb11 = bias;
+      float q00 = bias, q01 = bias, q10 = bias, q11 = bias;
       #pragma unroll
       for (int c = 0; c < 10; ++c) {
-        const float* pp = p1 + c * 144 + ph * 2 * 12 + pw0 * 2;
+        const float* pp = p1 + c * 144 + ph * 2 * 12 + pw * 2;
         const float* wc = w2s + (k * 10 + c) * 25;
         #pragma unroll
         for (int r = 0; r < 5; ++r) {
-          const float* rowx = pp + r * 12;
-          const float* rowy = rowx + 12;
-          float x0 = rowx[0], x1 = rowx[1], x2 = rowx[2], x3 = rowx[3];
-          float x4 = rowx[4], x5 = rowx[5], x6 = rowx[6], x7 = rowx[7];
-          float y0 = rowy[0], y1 = rowy[1], y2 = rowy[2], y3 = rowy[3];
-          float y4 = rowy[4], y5 = rowy[5], y6 = rowy[6], y7 = rowy[7];
-          const float xv[8] = {x0, x1, x2, x3, x4, x5, x6, x7};
-          const float yv[8] = {y0, y1, y2, y3, y4, y5, y6, y7};
           #pragma unroll
           for (int s = 0; s < 5; ++s) {
             const float w = wc[r * 5 + s];
-            qa00 += xv[s] * w;
-            qa01 += xv[s + 1] * w;
-            qa10 += yv[s] * w;
-            qa11 += yv[s + 1] * w;
-            qb00 += xv[s + 2] * w;
-            qb01 += xv[s + 3] * w;
-            qb10 += yv[s + 2] * w;
-            qb11 += yv[s + 3] * w;
+            const float* pr = pp + r * 12 + s;
+            q00 += pr[0] * w;
+            q01 += pr[1] * w;
+            q10 += pr[12] * w;
+            q11 += pr[13] * w;
           }
         }
       }
-      float dsc = 1.f;
       if (training) {
         const uint32_t rr = mix32(seed, (uint64_t)b * N_C2K + k);
-        dsc = (rr >= 0x80000000u) ? 2.f : 0.f;
+        const float dsc = (rr >= 0x80000000u) ? 2.f : 0.f;
+        q00 *= dsc; q01 *= dsc; q10 *= dsc; q11 *= dsc;
       }
-      #pragma unroll
-      for (int cell = 0; cell < 2; ++cell) {
-        const float q00 = (cell ? qb00 : qa00) * dsc;
-        const float q01 = (cell ? qb01 : qa01) * dsc;
-        const float q10 = (cell ? qb10 : qa10) * dsc;
-        const float q11 = (cell ? qb11 : qa11) * dsc;
-        int am = 0; float m = q00;
-        if (q01 > m) { m = q01; am = 1; }
-        if (q10 > m) { m = q10; am = 2; }
-        if (q11 > m) { m = q11; am = 3; }
-        const float o = m > 0.f ? m : 0.f;
-        const int ii = k * 16 + ph * 4 + pw0 + cell;
-        p2[ii] = o;
-        p2_ws[(int64_t)b * N_P2 + ii] = o;
-        idx2_ws[(int64_t)b * N_P2 + ii] =
-            (uint8_t)(m > 0.f ? am : (am | 4));
-      }
+      int am = 0; float m = q00;
+      if (q01 > m) { m = q01; am = 1; }
+      if (q10 > m) { m = q10; am = 2; }
+      if (q11 > m) { m = q11; am = 3; }
+      const float o = m > 0.f ? m : 0.f;
+      p2[i] = o;
+      p2_ws[(int64_t)b * N_P2 + i] = o;
+      idx2_ws[(int64_t)b * N_P2 + i] = (uint8_t)(m > 0.f ? am : (am | 4));
     }
     if (training && tid < N_C2K) {
       const uint32_t rr = mix32(seed, (uint64_t)b * N_C2K + tid);
