@@ -1511,6 +1511,10 @@ __device__ __forceinline__ void net_gw_tile(
       if (it < n_w) {
         const int e = it / rows, oh = oh0 + it % rows;
         const int k = e / 25, r = (e / 5) % 5, sx = e % 5;
+        // unroll 2: each sample's rows are a cold HBM first-touch for
+        // this block; interleaving two iterations overlaps the load
+        // latency the serial accumulate chain was exposing (r2)
+        #pragma unroll 2
         for (int b = b0; b < b1; ++b) {
           const float* grow = ga1_ws + (int64_t)b * N_A1 + k * 576 +
                               oh * 24;
