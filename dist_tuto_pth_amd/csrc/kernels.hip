@@ -1260,7 +1260,7 @@ net_fused_bwd_kernel(
 // (grid = B, not B*split), so at small B the chip is underfilled.
 // The backward LDS carve aliases the forward's dead buffers: total
 // 5020 (w2s, shared by both phases — same [k*10+c]*25 weight layout)
-// + 6848 floats = 46.5 KB/WG -> 3 WGs/CU.
+// + 7168 floats = 47.7 KB/WG -> 3 WGs/CU.
 __global__ void
 __launch_bounds__(256)
 net_fused_fwdbwd_kernel(
