@@ -32,7 +32,7 @@ plumbing parent<->child on one device.
 
 from __future__ import annotations
 
-from typing import Dict, Optional, Tuple
+from typing import Dict, Optional
 
 import torch
 

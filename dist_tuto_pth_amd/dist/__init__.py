@@ -28,7 +28,6 @@ from __future__ import annotations
 
 import os
 import socket
-import threading
 import time
 from typing import List, Optional, Sequence
 

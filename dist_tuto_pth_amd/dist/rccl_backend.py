@@ -16,7 +16,6 @@ on a GPU machine — there is no silent eager fallback.
 from __future__ import annotations
 
 import os
-from typing import Optional
 
 import torch
 

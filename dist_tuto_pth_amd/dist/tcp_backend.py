@@ -26,7 +26,6 @@ backend remains available separately as a cross-check.
 from __future__ import annotations
 
 import threading
-from typing import Optional
 
 import torch
 
