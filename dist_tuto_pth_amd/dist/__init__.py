@@ -11,8 +11,11 @@ owned by this package:
     (``csrc/rcclx.cpp``) driving RCCL over xGMI, one rank per GPU,
     rendezvous through a C++ TCP store (the master/worker handshake the
     tutorial describes at tuto.md:409-419, rebuilt natively).
-  * backend ``"gloo"`` — CPU plumbing for tests and the no-GPU ptp config
-    (BASELINE.md config 1), delegated to torch.distributed's gloo.
+  * backend ``"tcp"`` — the reference's CPU THD channel, owned natively:
+    a C++ per-pair socket mesh with collectives composed from p2p
+    (tcp_backend.py).
+  * backend ``"gloo"`` — torch.distributed delegation, kept as a CPU
+    cross-check.
 
 Exposed (modern forms of the tutorial's 0.x API):
 ``init_process_group``, ``get_rank``, ``get_world_size``, ``new_group``,
@@ -217,7 +220,8 @@ def init_process_group(
     backend:
       * ``"rccl"`` (alias ``"nccl"``): native RCCL-over-xGMI backend,
         one rank per MI355X.  Device defaults to ``rank % device_count``.
-      * ``"gloo"``: CPU plumbing path (tests, BASELINE config 1).
+      * ``"tcp"``: the native CPU socket-mesh backend (config 1).
+      * ``"gloo"``: torch.distributed delegation (CPU cross-check).
     """
     if _state.backend is not None:
         raise RuntimeError("process group already initialized")
