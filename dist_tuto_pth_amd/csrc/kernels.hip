@@ -2180,7 +2180,16 @@ static int gw_nch(int B) {
     const char* e = std::getenv("DTP_GW_NCH");
     env_nch = e ? std::atoi(e) : -1;
   }
-  int nch = (env_nch > 0) ? env_nch : 32;
+  int nch;
+  if (env_nch > 0) {
+    nch = env_nch;
+  } else if (B <= 192) {
+    nch = 16;   // B=128: 1.67M vs 1.61M at 32 (r2 sweep, post-rework)
+  } else if (B <= 768) {
+    nch = 24;   // B=512: 3.31M vs 3.06M at 32
+  } else {
+    nch = 32;   // B>=1024: 32 still best (B=4096: 4.76M vs 4.61M at 24)
+  }
   if (nch > 32) nch = 32;  // the part workspace holds 32 rows (_ws)
   if (nch > B) nch = B;
   return nch;
