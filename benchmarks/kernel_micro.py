@@ -90,9 +90,12 @@ def main():
     bwd()
     torch.cuda.synchronize()
 
-    bchunk = (B + 31) // 32
-    nch = (B + bchunk - 1) // bchunk
-    c1_ext = (8 if bchunk <= 16 else 24) - 1  # keep in sync: gw_c1_subs
+    # mirror csrc gw_nch()/gw_c1_subs() so raw segment launches match
+    # what the fused step actually runs
+    nch = int(os.environ.get("DTP_GW_NCH", 0)) or         (16 if B <= 192 else (24 if B <= 768 else 32))
+    nch = min(nch, 32, B)
+    bchunk = (B + nch - 1) // nch
+    c1_ext = (8 if bchunk <= 16 else 24) - 1
 
     def seg(base, ntiles):
         def f():
