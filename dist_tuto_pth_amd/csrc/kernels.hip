@@ -1347,7 +1347,31 @@ static_assert(T_CONV2 == N_C2K, "conv2 gw tiling: one tile per output "
 // r), 200 lanes, owns all FIVE sx weights of its (c, r) row: one
 // 12-value xr window + one 8-value gr row feed 40 FMA per channel.
 // Staging is double-buffered float4.
-template <int NK>
+// partial-row access, parameterized on coherence: the in-launch fold
+// variant stores partials sc1 (device-coherent, straight to the
+// coherence point) and reads them back sc1 — ZERO fences, so the
+// concurrently-running tile blocks keep their L1/L2 activation
+// locality.  The first fold attempt used the plain-store + agent
+// release/acquire recipe and measured 23-28% SLOWER end to end: one
+// `buffer_wbl2` L2 writeback per block x 1744 blocks evicted the very
+// activations the partial blocks re-read (ledger).  Partial traffic is
+// ~1.8 MB/step, so the slower sc1 store path is immaterial.
+template <bool SC1>
+__device__ __forceinline__ void gw_st(float* p, float v) {
+  if (SC1)
+    __hip_atomic_store(p, v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+  else
+    *p = v;
+}
+template <bool SC1>
+__device__ __forceinline__ float gw_ld(const float* p) {
+  if (SC1)
+    return __hip_atomic_load(p, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_AGENT);
+  return *p;
+}
+
+template <int NK, bool SC1 = false>
 __device__ __forceinline__ void net_gw_conv2_fold(
     int k0, int tid, int b0, int b1, float* __restrict__ my,
     const float* __restrict__ p1_ws, const float* __restrict__ ga2_ws) {
@@ -1443,12 +1467,12 @@ __device__ __forceinline__ void net_gw_conv2_fold(
   if (tid < 250) {
     #pragma unroll
     for (int kk = 0; kk < NK; ++kk)
-      my[OFF_W2 + (k0 + kk) * 250 + tid] = wacc[kk][tid];
+      gw_st<SC1>(my + OFF_W2 + (k0 + kk) * 250 + tid, wacc[kk][tid]);
   }
   if (tid == 250) {
     #pragma unroll
     for (int kk = 0; kk < NK; ++kk)
-      my[OFF_B2 + k0 + kk] = wacc[kk][250];
+      gw_st<SC1>(my + OFF_B2 + k0 + kk, wacc[kk][250]);
   }
 }
 
@@ -1456,6 +1480,7 @@ __device__ __forceinline__ void net_gw_conv2_fold(
 // walk [conv2 | fc1 | conv1x4 | fc2]; partials land in my[GW_ROW]
 // laid out like the flat grad buffer.  Shared by net_gw_partial_kernel
 // and the single-launch net_step_kernel.
+template <bool SC1 = false>
 __device__ __forceinline__ void net_gw_tile(
     int tile, int tid, int b0, int b1, int c1_subs,
     float* __restrict__ my,
@@ -1478,9 +1503,10 @@ __device__ __forceinline__ void net_gw_tile(
     //    which beats the re-read saving end to end.
     if (b1 - b0 <= 16) {
       if (tile < N_C2K / 2)
-        net_gw_conv2_fold<2>(tile * 2, tid, b0, b1, my, p1_ws, ga2_ws);
+        net_gw_conv2_fold<2, SC1>(tile * 2, tid, b0, b1, my, p1_ws,
+                                  ga2_ws);
     } else {
-      net_gw_conv2_fold<1>(tile, tid, b0, b1, my, p1_ws, ga2_ws);
+      net_gw_conv2_fold<1, SC1>(tile, tid, b0, b1, my, p1_ws, ga2_ws);
     }
     return;
   }
@@ -1542,7 +1568,7 @@ __device__ __forceinline__ void net_gw_tile(
     }
     __syncthreads();
     float* dst = (sub == 0) ? (my + OFF_W1) : (my + GW_TOTAL + (sub - 1) * 260);
-    for (int i = tid; i < 260; i += 256) dst[i] = wacc[i];
+    for (int i = tid; i < 260; i += 256) gw_st<SC1>(dst + i, wacc[i]);
     return;
   }
   tile -= T_CONV1;
@@ -1564,12 +1590,12 @@ __device__ __forceinline__ void net_gw_tile(
           ae += gh1_ws[(int64_t)b * N_H1 + n] *
                 p2_ws[(int64_t)b * N_P2 + k];
         acc = ae + ao;
-        my[OFF_WF1 + i] = acc;
+        gw_st<SC1>(my + OFF_WF1 + i, acc);
       } else {
         const int n = i - 16000;
         for (int b = b0; b < b1; ++b)
           acc += gh1_ws[(int64_t)b * N_H1 + n];
-        my[OFF_BF1 + n] = acc;
+        gw_st<SC1>(my + OFF_BF1 + n, acc);
       }
     }
     return;
@@ -1584,12 +1610,12 @@ __device__ __forceinline__ void net_gw_tile(
         for (int b = b0; b < b1; ++b)
           acc += glog_ws[(int64_t)b * N_CLS + n] *
                  d3_ws[(int64_t)b * N_H1 + k];
-        my[OFF_WF2 + i] = acc;
+        gw_st<SC1>(my + OFF_WF2 + i, acc);
       } else {
         const int n = i - 500;
         for (int b = b0; b < b1; ++b)
           acc += glog_ws[(int64_t)b * N_CLS + n];
-        my[OFF_BF2 + n] = acc;
+        gw_st<SC1>(my + OFF_BF2 + n, acc);
       }
     }
   }
@@ -1617,6 +1643,7 @@ net_gw_partial_kernel(int c1_subs, const float* __restrict__ x,
 // the 8 per-parameter pointers (which may alias one flat buffer).
 struct GwPtrs { float* p[8]; };
 // sum flat-grad element i over the nch chunk rows (+ conv1 extension)
+template <bool SC1 = false>
 __device__ __forceinline__ float net_gw_combine_elem(
     int i, int nch, int c1_ext, const float* __restrict__ part) {
   // four independent accumulator chains: the single 32-deep
@@ -1624,12 +1651,12 @@ __device__ __forceinline__ float net_gw_combine_elem(
   float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
   int c = 0;
   for (; c + 3 < nch; c += 4) {
-    a0 += part[(int64_t)c * GW_ROW + i];
-    a1 += part[(int64_t)(c + 1) * GW_ROW + i];
-    a2 += part[(int64_t)(c + 2) * GW_ROW + i];
-    a3 += part[(int64_t)(c + 3) * GW_ROW + i];
+    a0 += gw_ld<SC1>(part + (int64_t)c * GW_ROW + i);
+    a1 += gw_ld<SC1>(part + (int64_t)(c + 1) * GW_ROW + i);
+    a2 += gw_ld<SC1>(part + (int64_t)(c + 2) * GW_ROW + i);
+    a3 += gw_ld<SC1>(part + (int64_t)(c + 3) * GW_ROW + i);
   }
-  for (; c < nch; ++c) a0 += part[(int64_t)c * GW_ROW + i];
+  for (; c < nch; ++c) a0 += gw_ld<SC1>(part + (int64_t)c * GW_ROW + i);
   if (i < 260) {  // conv1 sub-block extension rows (see GW_ROW)
     // c1_ext is 7 (band mode) or 23 (row mode): branch to fully
     // unrolled folds — a runtime-bound loop here cost the combine
@@ -1637,20 +1664,21 @@ __device__ __forceinline__ float net_gw_combine_elem(
     if (c1_ext == 7) {
       for (int c2 = 0; c2 < nch; ++c2) {
         const float* ext = part + (int64_t)c2 * GW_ROW + GW_TOTAL;
-        a0 += ext[i] + ext[4 * 260 + i];
-        a1 += ext[260 + i] + ext[5 * 260 + i];
-        a2 += ext[2 * 260 + i] + ext[6 * 260 + i];
-        a3 += ext[3 * 260 + i];
+        a0 += gw_ld<SC1>(ext + i) + gw_ld<SC1>(ext + 4 * 260 + i);
+        a1 += gw_ld<SC1>(ext + 260 + i) + gw_ld<SC1>(ext + 5 * 260 + i);
+        a2 += gw_ld<SC1>(ext + 2 * 260 + i) +
+              gw_ld<SC1>(ext + 6 * 260 + i);
+        a3 += gw_ld<SC1>(ext + 3 * 260 + i);
       }
     } else {
       for (int c2 = 0; c2 < nch; ++c2) {
         const float* ext = part + (int64_t)c2 * GW_ROW + GW_TOTAL;
         #pragma unroll
         for (int s = 0; s < 23; s += 4) {
-          a0 += ext[s * 260 + i];
-          if (s + 1 < 23) a1 += ext[(s + 1) * 260 + i];
-          if (s + 2 < 23) a2 += ext[(s + 2) * 260 + i];
-          if (s + 3 < 23) a3 += ext[(s + 3) * 260 + i];
+          a0 += gw_ld<SC1>(ext + s * 260 + i);
+          if (s + 1 < 23) a1 += gw_ld<SC1>(ext + (s + 1) * 260 + i);
+          if (s + 2 < 23) a2 += gw_ld<SC1>(ext + (s + 2) * 260 + i);
+          if (s + 3 < 23) a3 += gw_ld<SC1>(ext + (s + 3) * 260 + i);
         }
       }
     }
@@ -1788,6 +1816,140 @@ __global__ void net_gw_combine_sgd_kernel(const float* __restrict__ part,
   }
   if ((loss_part || seed_bump) && blockIdx.x == 0)
     net_loss_finalize(loss_part, loss_out, nblk_fwd, seed_bump);
+}
+
+// ---------------------------------------------------------------------------
+// In-launch gw fold (r2): the combine dispatch is pure latency (~10 us at
+// B=128 — 86 tiny blocks, an L2-chain fold of 1.4 MB of partials), so this
+// variant of the partial kernel deletes it: the LAST-ARRIVING block of each
+// tile column folds the region its column wrote (and, DO_SGD, applies the
+// momentum update), using the split-K last-arriver recipe from the CDNA
+// guide (G16): every block drains vmcnt + __syncthreads, lane 0 issues an
+// agent-scope RELEASE fence, restates the vmcnt wait, THEN takes a relaxed
+// agent-scope ticket; the block that draws target-1 acquires (agent) and
+// reads every row with plain loads.  Fold regions are per COLUMN (what the
+// column's own blocks wrote), so counter==target implies the region's rows
+// are all visible:
+//   conv2 column k  -> channel k (pair mode: column k<10 -> channels 2k,
+//                      2k+1; columns 10..19 wrote nothing and own nothing)
+//   conv1 family    -> ONE counter over all T_CONV1*nch blocks (its
+//                      sub-blocks share the 260 outputs via extension rows,
+//                      so no single column's completion suffices)
+//   fc1/fc2 column j -> its 256-element flat slice
+// Each fold runs as soon as ITS column's blocks are done — overlapped with
+// the other families still computing — so the only exposed cost is the
+// straggler family's own (tiny) fold.
+__device__ __forceinline__ void net_gw_fold_range(
+    int i0, int i1, int nch, int c1_ext, const float* __restrict__ part,
+    const GwPtrs& g, const GwPtrs& prm, const GwPtrs& buf, float lr,
+    float mu, bool do_sgd) {
+  const int off[9] = {OFF_W1, OFF_B1, OFF_W2, OFF_B2, OFF_WF1, OFF_BF1,
+                      OFF_WF2, OFF_BF2, GW_TOTAL};
+  for (int i = i0 + (int)threadIdx.x; i < i1; i += 256) {
+    const float acc = net_gw_combine_elem<true>(i, nch, c1_ext, part);
+    const int t = net_gw_tensor_of(i, off);
+    const int64_t j = i - off[t];
+    g.p[t][j] = acc;
+    if (do_sgd) {
+      float v = acc;
+      if (buf.p[t]) {
+        v = mu * buf.p[t][j] + acc;
+        buf.p[t][j] = v;
+      }
+      prm.p[t][j] -= lr * v;
+    }
+  }
+}
+
+template <bool DO_SGD>
+__global__ void __launch_bounds__(256)
+net_gw_partial_fold_kernel(int c1_subs, const float* __restrict__ x,
+                           const float* __restrict__ p1_ws,
+                           const float* __restrict__ p2_ws,
+                           const float* __restrict__ d3_ws,
+                           const float* __restrict__ ga1_ws,
+                           const float* __restrict__ ga2_ws,
+                           const float* __restrict__ gh1_ws,
+                           const float* __restrict__ glog_ws,
+                           float* __restrict__ part, int B, int bchunk,
+                           GwPtrs g, GwPtrs prm, GwPtrs buf, float lr,
+                           float mu, const float* __restrict__ loss_part,
+                           float* __restrict__ loss_out, int nblk_fwd,
+                           unsigned long long* seed_bump,
+                           unsigned int* __restrict__ cnt) {
+  const int b0 = blockIdx.y * bchunk;
+  net_gw_tile<true>(blockIdx.x, threadIdx.x, b0, min(B, b0 + bchunk),
+                    c1_subs, part + (int64_t)blockIdx.y * GW_ROW,
+                    x, p1_ws, p2_ws, d3_ws, ga1_ws, ga2_ws, gh1_ws,
+                    glog_ws);
+  // ---- last-arriver ticket, FENCE-FREE: the tile wrote its partials
+  // with sc1 stores (already at the coherence point once vmcnt
+  // retires), so the ticket needs only the vmcnt drain before it and
+  // the reducer reads the rows back with sc1 loads.  No buffer_wbl2 /
+  // buffer_inv anywhere — the concurrent tile blocks keep their
+  // L1/L2-resident activations (the fenced variant cost 23-28%).
+  const int nch = (int)gridDim.y;
+  const int xcol = (int)blockIdx.x;
+  const bool conv1_fam = (xcol >= T_CONV2 && xcol < T_CONV2 + T_CONV1);
+  const int ci = conv1_fam ? T_CONV2 : xcol;
+  const unsigned target =
+      conv1_fam ? (unsigned)(T_CONV1 * nch) : (unsigned)nch;
+  __shared__ unsigned s_tk;
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  if (threadIdx.x == 0)
+    s_tk = __hip_atomic_fetch_add(&cnt[ci], 1u, __ATOMIC_RELAXED,
+                                  __HIP_MEMORY_SCOPE_AGENT);
+  __syncthreads();
+  if (s_tk != target - 1) return;  // not the last arriver of this region
+  if (threadIdx.x == 0)
+    __hip_atomic_store(&cnt[ci], 0u, __ATOMIC_RELAXED,
+                       __HIP_MEMORY_SCOPE_AGENT);  // re-arm for next step
+  __syncthreads();
+  const int c1_ext = c1_subs - 1;
+  if (xcol < T_CONV2) {  // conv2: this column's channel(s)
+    int k0, nk;
+    if (bchunk <= 16) {  // pair mode (mirrors net_gw_tile's split)
+      if (xcol >= N_C2K / 2) return;  // idle pair columns wrote nothing
+      k0 = xcol * 2;
+      nk = 2;
+    } else {
+      k0 = xcol;
+      nk = 1;
+    }
+    net_gw_fold_range(OFF_W2 + k0 * 250, OFF_W2 + (k0 + nk) * 250, nch,
+                      c1_ext, part, g, prm, buf, lr, mu, DO_SGD);
+    net_gw_fold_range(OFF_B2 + k0, OFF_B2 + k0 + nk, nch, c1_ext, part,
+                      g, prm, buf, lr, mu, DO_SGD);
+    return;
+  }
+  if (conv1_fam) {  // conv1: canonical 260 + extension rows (c1_ext)
+    net_gw_fold_range(0, 260, nch, c1_ext, part, g, prm, buf, lr, mu,
+                      DO_SGD);
+    return;
+  }
+  if (xcol < T_CONV2 + T_CONV1 + T_FC1) {  // fc1 column slice
+    const int i0 = OFF_WF1 + (xcol - (T_CONV2 + T_CONV1)) * 256;
+    net_gw_fold_range(i0, min(OFF_WF1 + 16050, i0 + 256), nch, c1_ext,
+                      part, g, prm, buf, lr, mu, DO_SGD);
+    return;
+  }
+  // fc2 column slice; the last column also finalizes the loss partials
+  // and advances the dropout seed (wave 0 only — no extra LDS).
+  const int i0 = OFF_WF2 + (xcol - (T_CONV2 + T_CONV1 + T_FC1)) * 256;
+  net_gw_fold_range(i0, min(OFF_WF2 + 510, i0 + 256), nch, c1_ext, part,
+                    g, prm, buf, lr, mu, DO_SGD);
+  if (xcol == GW_TILES - 1 && threadIdx.x < 64) {
+    if (loss_part) {
+      float v = 0.f;
+      for (int i = threadIdx.x; i < nblk_fwd; i += 64) v += loss_part[i];
+      #pragma unroll
+      for (int s = 32; s > 0; s >>= 1) v += __shfl_down(v, s, 64);
+      if (threadIdx.x == 0 && loss_out) *loss_out = v;
+    }
+    if (threadIdx.x == 0 && seed_bump)
+      *seed_bump += 0x9E3779B97F4A7C15ull;
+  }
 }
 
 // ===========================================================================
@@ -2195,6 +2357,37 @@ static int gw_nch(int B) {
   return nch;
 }
 
+// in-launch gw fold (last-arriver epilogue in the partial kernel, no
+// combine dispatch).  DTP_GW_FOLD=1 opts in; the default stays the
+// two-kernel path — MEASURED NEGATIVE on both step shapes (ledger):
+// the combine's dispatch latency is hidden by the stream pipeline
+// (kernels enqueue back-to-back), so only its ~5 us execution is
+// exposed, while the epilogue's per-block sc1-drain tail costs ~4-5 us
+// across the critical path at B=128 (-5%); parity at B>=512.
+static bool gw_fold_on() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = std::getenv("DTP_GW_FOLD");
+    v = (e && std::atoi(e) != 0) ? 1 : 0;
+  }
+  return v == 1;
+}
+
+// per-device ticket counters for the fold epilogue (GW_TILES u32,
+// zeroed once; each reducer re-arms its own counter every step)
+static unsigned int* gw_cnt_buf(hipStream_t s) {
+  static unsigned int* bufs[64] = {};
+  int dev = 0;
+  HIP_CHECK(hipGetDevice(&dev));
+  if (dev < 0 || dev >= 64) throw std::runtime_error("device index > 64");
+  if (!bufs[dev]) {
+    HIP_CHECK(hipMalloc(&bufs[dev], GW_TILES * sizeof(unsigned int)));
+    HIP_CHECK(hipMemsetAsync(bufs[dev], 0, GW_TILES * sizeof(unsigned int),
+                             s));
+  }
+  return bufs[dev];
+}
+
 void net_fused_fwd(uintptr_t x, uintptr_t w1, uintptr_t b1, uintptr_t w2,
                    uintptr_t b2, uintptr_t wf1, uintptr_t bf1,
                    uintptr_t wf2, uintptr_t bf2, uintptr_t tgt,
@@ -2323,6 +2516,27 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
   // combine, more batch per partial block — sweep on hardware).
   const int nch = gw_nch(B);
   const int bchunk = (B + nch - 1) / nch;
+  GwPtrs gp;
+  gp.p[0] = (float*)gw1; gp.p[1] = (float*)gb1;
+  gp.p[2] = (float*)gw2; gp.p[3] = (float*)gb2;
+  gp.p[4] = (float*)gwf1; gp.p[5] = (float*)gbf1;
+  gp.p[6] = (float*)gwf2; gp.p[7] = (float*)gbf2;
+  unsigned long long* sb = (training && seed_dev)
+      ? (unsigned long long*)seed_dev : nullptr;
+  if (gw_fold_on()) {
+    GwPtrs none{};
+    hipLaunchKernelGGL(net_gw_partial_fold_kernel<false>,
+                       dim3(GW_TILES, nch), dim3(256), 0, S(stream),
+                       gw_c1_subs(bchunk), (const float*)x,
+                       (const float*)p1_ws, (const float*)p2_ws,
+                       (const float*)d3_ws, (const float*)ga1_ws,
+                       (const float*)ga2_ws, (const float*)gh1_ws,
+                       (const float*)glog_ws, (float*)part_ws, B, bchunk,
+                       gp, none, none, 0.f, 0.f,
+                       (const float*)loss_part, (float*)loss_out,
+                       grid_for(B, 1), sb, gw_cnt_buf(S(stream)));
+    return;
+  }
   hipLaunchKernelGGL(net_gw_partial_kernel, dim3(GW_TILES, nch),
                      dim3(256), 0, S(stream), gw_c1_subs(bchunk),
                      (const float*)x,
@@ -2331,19 +2545,12 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
                      (const float*)ga2_ws, (const float*)gh1_ws,
                      (const float*)glog_ws, (float*)part_ws, B, bchunk,
                      0);
-  GwPtrs gp;
-  gp.p[0] = (float*)gw1; gp.p[1] = (float*)gb1;
-  gp.p[2] = (float*)gw2; gp.p[3] = (float*)gb2;
-  gp.p[4] = (float*)gwf1; gp.p[5] = (float*)gbf1;
-  gp.p[6] = (float*)gwf2; gp.p[7] = (float*)gbf2;
   hipLaunchKernelGGL(net_gw_combine_kernel,
                      dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
                      S(stream), (const float*)part_ws, gp, nch,
                      gw_c1_subs(bchunk) - 1,
                      (const float*)loss_part, (float*)loss_out,
-                     grid_for(B, 1),
-                     (training && seed_dev)
-                         ? (unsigned long long*)seed_dev : nullptr);
+                     grid_for(B, 1), sb);
 }
 
 // Combined fwd+bwd (one dispatch) + gw partial + combine: a 3-dispatch
@@ -2382,6 +2589,31 @@ void net_fused_fwdbwd(
                      training ? 1 : 0);
   const int nch = gw_nch(B);
   const int bchunk = (B + nch - 1) / nch;
+  GwPtrs gp{}, pp{}, bp{};
+  for (int i = 0; i < 8; ++i) gp.p[i] = (float*)grd_v[i];
+  const bool sgd = !prm_v.empty();
+  if (sgd) {
+    for (int i = 0; i < 8; ++i) {
+      pp.p[i] = (float*)prm_v[i];
+      bp.p[i] = (i < (int)buf_v.size()) ? (float*)buf_v[i] : nullptr;
+    }
+  }
+  unsigned long long* sb = (training && seed_dev)
+      ? (unsigned long long*)seed_dev : nullptr;
+  if (gw_fold_on()) {
+    auto* kern = sgd ? net_gw_partial_fold_kernel<true>
+                     : net_gw_partial_fold_kernel<false>;
+    hipLaunchKernelGGL(kern, dim3(GW_TILES, nch), dim3(256), 0, S(stream),
+                       gw_c1_subs(bchunk), (const float*)x,
+                       (const float*)p1_ws, (const float*)p2_ws,
+                       (const float*)d3_ws, (const float*)ga1_ws,
+                       (const float*)ga2_ws, (const float*)gh1_ws,
+                       (const float*)glog_ws, (float*)part_ws, B, bchunk,
+                       gp, pp, bp, (float)lr, (float)mu,
+                       (const float*)loss_part, (float*)loss_out, nblk,
+                       sb, gw_cnt_buf(S(stream)));
+    return;
+  }
   hipLaunchKernelGGL(net_gw_partial_kernel, dim3(GW_TILES, nch),
                      dim3(256), 0, S(stream), gw_c1_subs(bchunk),
                      (const float*)x,
@@ -2390,11 +2622,7 @@ void net_fused_fwdbwd(
                      (const float*)ga2_ws, (const float*)gh1_ws,
                      (const float*)glog_ws, (float*)part_ws, B, bchunk,
                      0);
-  GwPtrs gp{};
-  for (int i = 0; i < 8; ++i) gp.p[i] = (float*)grd_v[i];
-  unsigned long long* sb = (training && seed_dev)
-      ? (unsigned long long*)seed_dev : nullptr;
-  if (prm_v.empty()) {
+  if (!sgd) {
     hipLaunchKernelGGL(net_gw_combine_kernel,
                        dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
                        S(stream), (const float*)part_ws, gp, nch,
@@ -2402,11 +2630,6 @@ void net_fused_fwdbwd(
                        (const float*)loss_part, (float*)loss_out,
                        nblk, sb);
   } else {
-    GwPtrs pp{}, bp{};
-    for (int i = 0; i < 8; ++i) {
-      pp.p[i] = (float*)prm_v[i];
-      bp.p[i] = (i < (int)buf_v.size()) ? (float*)buf_v[i] : nullptr;
-    }
     hipLaunchKernelGGL(net_gw_combine_sgd_kernel,
                        dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
                        S(stream), (const float*)part_ws, gp, pp, bp, nch,
@@ -2454,6 +2677,27 @@ void net_fused_bwd_sgd(uintptr_t x, uintptr_t w2, uintptr_t wf1,
                      training ? 1 : 0, split);
   const int nch = gw_nch(B);
   const int bchunk = (B + nch - 1) / nch;
+  GwPtrs gp{}, pp{}, bp{};
+  for (int i = 0; i < 8; ++i) {
+    gp.p[i] = (float*)grd_v[i];
+    pp.p[i] = (float*)prm_v[i];
+    bp.p[i] = (i < (int)buf_v.size()) ? (float*)buf_v[i] : nullptr;
+  }
+  unsigned long long* sb = (training && seed_dev)
+      ? (unsigned long long*)seed_dev : nullptr;
+  if (gw_fold_on()) {
+    hipLaunchKernelGGL(net_gw_partial_fold_kernel<true>,
+                       dim3(GW_TILES, nch), dim3(256), 0, S(stream),
+                       gw_c1_subs(bchunk), (const float*)x,
+                       (const float*)p1_ws, (const float*)p2_ws,
+                       (const float*)d3_ws, (const float*)ga1_ws,
+                       (const float*)ga2_ws, (const float*)gh1_ws,
+                       (const float*)glog_ws, (float*)part_ws, B, bchunk,
+                       gp, pp, bp, (float)lr, (float)mu,
+                       (const float*)loss_part, (float*)loss_out,
+                       grid_for(B, 1), sb, gw_cnt_buf(S(stream)));
+    return;
+  }
   hipLaunchKernelGGL(net_gw_partial_kernel, dim3(GW_TILES, nch),
                      dim3(256), 0, S(stream), gw_c1_subs(bchunk),
                      (const float*)x,
@@ -2462,20 +2706,12 @@ void net_fused_bwd_sgd(uintptr_t x, uintptr_t w2, uintptr_t wf1,
                      (const float*)ga2_ws, (const float*)gh1_ws,
                      (const float*)glog_ws, (float*)part_ws, B, bchunk,
                      0);
-  GwPtrs gp{}, pp{}, bp{};
-  for (int i = 0; i < 8; ++i) {
-    gp.p[i] = (float*)grd_v[i];
-    pp.p[i] = (float*)prm_v[i];
-    bp.p[i] = (i < (int)buf_v.size()) ? (float*)buf_v[i] : nullptr;
-  }
   hipLaunchKernelGGL(net_gw_combine_sgd_kernel,
                      dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
                      S(stream), (const float*)part_ws, gp, pp, bp, nch,
                      gw_c1_subs(bchunk) - 1,
                      (float)lr, (float)mu, (const float*)loss_part,
-                     (float*)loss_out, grid_for(B, 1),
-                     (training && seed_dev)
-                         ? (unsigned long long*)seed_dev : nullptr);
+                     (float*)loss_out, grid_for(B, 1), sb);
 }
 
 // raw combine launch (microbenchmarks: time the combine/sgd dispatch
