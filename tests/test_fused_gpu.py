@@ -297,3 +297,46 @@ def test_fwdbwd_training_convergence():
     first = sum(losses[:10]) / 10
     last = sum(losses[-10:]) / 10
     assert last < first - 0.12, (first, last)
+
+
+def test_fold_optin_matches_default_subprocess():
+    """The opt-in in-launch gw fold (DTP_GW_FOLD=1, a documented
+    measured-negative kept for the record — see
+    profiles/convnet_step_history.md) must stay numerically correct:
+    run one fused step in a subprocess with the flag set (it is read
+    once per process) and compare grads against this process's
+    default combine path."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    net_c, net_g, x, tgt = _mk(11)
+    net_fused_step(net_g, x.to(DEV), tgt.to(DEV))
+    torch.cuda.synchronize()
+    want = {n: p.grad.cpu() for n, p in net_g.named_parameters()}
+
+    prog = (
+        "import json, sys, torch\n"
+        "from dist_tuto_pth_amd.models import Net\n"
+        "from dist_tuto_pth_amd.ops.fused import net_fused_step\n"
+        "torch.manual_seed(11)\n"
+        "net_c = Net().eval(); net_g = Net().eval().to('cuda:0')\n"
+        "net_g.load_state_dict({k: v.to('cuda:0')\n"
+        "                       for k, v in net_c.state_dict().items()})\n"
+        "x = torch.randn(64, 1, 28, 28); tgt = torch.randint(0, 10, (64,))\n"
+        "net_fused_step(net_g, x.to('cuda:0'), tgt.to('cuda:0'))\n"
+        "torch.cuda.synchronize()\n"
+        "out = {n: p.grad.cpu().flatten().tolist()\n"
+        "       for n, p in net_g.named_parameters()}\n"
+        "print(json.dumps(out))\n"
+    )
+    env = dict(os.environ, DTP_GW_FOLD="1")
+    r = subprocess.run([sys.executable, "-c", prog], env=env,
+                       capture_output=True, text=True, timeout=180)
+    assert r.returncode == 0, r.stderr[-2000:]
+    got = json.loads(r.stdout.strip().splitlines()[-1])
+    for n, w in want.items():
+        g = torch.tensor(got[n]).view_as(w)
+        assert torch.allclose(g, w, atol=1e-6), \
+            (n, (g - w).abs().max().item())
