@@ -47,8 +47,14 @@ def parse_args():
     p.add_argument("--no-fused", action="store_true",
                    help="use the modular per-op kernel pipeline instead "
                         "of the fused whole-Net kernels")
-    p.add_argument("--fwdbwd", action="store_true",
-                   help="combined fwd+bwd kernel (3-dispatch step)")
+    p.add_argument("--fwdbwd", dest="fwdbwd", action="store_true",
+                   default=None,
+                   help="combined fwd+bwd kernel (3-dispatch step); "
+                        "default: auto — on for 192 <= batch <= 512, "
+                        "where it measures +1..+4% (gpurun_out/"
+                        "fwdbwd_band sweep), off elsewhere (loses 8% "
+                        "at B=128, 4% at B=768)")
+    p.add_argument("--no-fwdbwd", dest="fwdbwd", action="store_false")
     p.add_argument("--megakernel", action="store_true",
                    help="run the whole step as ONE cooperative kernel "
                         "launch (measured slower than the 6-dispatch "
@@ -59,6 +65,8 @@ def parse_args():
 
 def main():
     args = parse_args()
+    if args.fwdbwd is None:
+        args.fwdbwd = 192 <= args.batch <= 512
     world = int(os.environ.get("WORLD_SIZE", args.gpus))
     rank = int(os.environ.get("RANK", 0))
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
