@@ -826,13 +826,50 @@ __global__ void copy_throttled_f32_kernel(float* __restrict__ dst,
 #define N_H1 50         // fc1 out
 #define N_CLS 10        // classes
 
+// partial-row access, parameterized on coherence: the in-launch fold
+// variant stores partials sc1 (device-coherent, straight to the
+// coherence point) and reads them back sc1 — ZERO fences, so the
+// concurrently-running tile blocks keep their L1/L2 activation
+// locality.  The first fold attempt used the plain-store + agent
+// release/acquire recipe and measured 23-28% SLOWER end to end: one
+// `buffer_wbl2` L2 writeback per block x 1744 blocks evicted the very
+// activations the partial blocks re-read (ledger).  Partial traffic is
+// ~1.8 MB/step, so the slower sc1 store path is immaterial.
+template <bool SC1>
+__device__ __forceinline__ void gw_st(float* p, float v) {
+  if (SC1)
+    __hip_atomic_store(p, v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+  else
+    *p = v;
+}
+template <bool SC1>
+__device__ __forceinline__ float gw_ld(const float* p) {
+  if (SC1)
+    return __hip_atomic_load(p, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_AGENT);
+  return *p;
+}
+
 // One sample's full forward (conv1..log_softmax+NLL) by one 256-thread
 // workgroup, activations staged through caller-provided LDS buffers.
 // Returns -— on tid 0 only — the sample's log-prob at the target
 // (callers accumulate the NLL loss); other lanes return 0.
 // Shared by net_fused_fwd_kernel and the single-launch net_step_kernel.
+// SPLIT mode (fused fwd at small B): TWO sibling workgroups per sample
+// fill the otherwise half-idle chip (B=128 launches 128 blocks on 256
+// CUs).  Both stage x/w1 and compute conv1+pool1 fully (duplicating
+// the cheap stage costs less than a second exchange), each computes
+// HALF of conv2's 20 output channels (and stages only its half of w2),
+// sibling 1 publishes its p2 half with sc1 stores + a per-sample flag
+// and exits; sibling 0 consumes it and runs fc1/fc2/softmax alone.
+// If the sibling is not yet resident (co-residency is not contractual)
+// sibling 0 falls back to computing the other half itself from its own
+// LDS p1 — identical values, so the duplicate global stores are benign
+// and the wait can never deadlock.
+template <bool SPLIT = false>
 __device__ __forceinline__ float net_fwd_sample(
     int b, int tid, int B, int training, uint64_t seed,
+    int half, unsigned int* __restrict__ flag,
     const float* __restrict__ x,
     const float* __restrict__ w1, const float* __restrict__ b1,
     const float* __restrict__ w2, const float* __restrict__ b2,
@@ -853,8 +890,14 @@ __device__ __forceinline__ float net_fwd_sample(
     for (int i = tid; i < 784; i += 256) xs[i] = x[(int64_t)b * 784 + i];
     for (int i = tid; i < N_C1K * 25; i += 256) w1s[i] = w1[i];
     if (tid < N_C1K) w1s[N_C1K * 25 + tid] = b1[tid];
-    for (int i = tid; i < N_C2K * 250; i += 256) w2s[i] = w2[i];
-    if (tid < N_C2K) w2s[N_C2K * 250 + tid] = b2[tid];
+    {
+      const int wlo = SPLIT ? half * (N_C2K / 2) * 250 : 0;
+      const int whi = SPLIT ? wlo + (N_C2K / 2) * 250 : N_C2K * 250;
+      for (int i = wlo + tid; i < whi; i += 256) w2s[i] = w2[i];
+      const int klo = SPLIT ? half * (N_C2K / 2) : 0;
+      const int khi = SPLIT ? klo + N_C2K / 2 : N_C2K;
+      if (klo + tid < khi) w2s[N_C2K * 250 + klo + tid] = b2[klo + tid];
+    }
     __syncthreads();
 
     // conv1 + pool1 + relu in ONE register-blocked stage: each thread
@@ -886,15 +929,20 @@ __device__ __forceinline__ float net_fwd_sample(
       if (q11 > m) { m = q11; am = 3; }
       const float o = m > 0.f ? m : 0.f;
       p1[i] = o;
-      p1_ws[(int64_t)b * N_P1 + i] = o;
-      idx1_ws[(int64_t)b * N_P1 + i] = (uint8_t)(m > 0.f ? am : (am | 4));
+      if (!SPLIT || half == 0) {
+        p1_ws[(int64_t)b * N_P1 + i] = o;
+        idx1_ws[(int64_t)b * N_P1 + i] =
+            (uint8_t)(m > 0.f ? am : (am | 4));
+      }
     }
     __syncthreads();
 
     // conv2 + dropout2d + pool2 + relu, same register-blocked shape:
     // one thread per pooled cell, four conv outputs in four chains,
     // channel dropout applied in registers before the max.
-    for (int i = tid; i < N_P2; i += 256) {
+    const int c2lo = SPLIT ? half * (N_P2 / 2) : 0;
+    const int c2hi = SPLIT ? c2lo + N_P2 / 2 : N_P2;
+    for (int i = c2lo + tid; i < c2hi; i += 256) {
       const int k = i / 16, ph = (i / 4) % 4, pw = i % 4;
       const float bias = w2s[N_C2K * 250 + k];
       float q00 = bias, q01 = bias, q10 = bias, q11 = bias;
@@ -926,14 +974,79 @@ __device__ __forceinline__ float net_fwd_sample(
       if (q11 > m) { m = q11; am = 3; }
       const float o = m > 0.f ? m : 0.f;
       p2[i] = o;
-      p2_ws[(int64_t)b * N_P2 + i] = o;
+      gw_st<SPLIT>(p2_ws + (int64_t)b * N_P2 + i, o);
       idx2_ws[(int64_t)b * N_P2 + i] = (uint8_t)(m > 0.f ? am : (am | 4));
     }
-    if (training && tid < N_C2K) {
+    if ((!SPLIT || half == 0) && training && tid < N_C2K) {
       const uint32_t rr = mix32(seed, (uint64_t)b * N_C2K + tid);
       m2_ws[(int64_t)b * N_C2K + tid] = rr >= 0x80000000u;
     }
     __syncthreads();
+    if (SPLIT && half == 1) {
+      // publish: p2 half is sc1-stored (at the coherence point once
+      // vmcnt retires) — drain, then raise the per-sample flag.
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+      if (tid == 0)
+        __hip_atomic_store(flag, 1u, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+      return 0.f;  // fc/softmax belong to sibling 0
+    }
+    if (SPLIT && half == 0) {
+      // consume the sibling's half: bounded poll, then sc1 loads into
+      // our LDS p2.  On timeout compute the half ourselves (fallback).
+      __shared__ unsigned s_got;
+      if (tid == 0) {
+        unsigned got = 0;
+        for (int it = 0; it < 60000; ++it) {
+          if (__hip_atomic_load(flag, __ATOMIC_RELAXED,
+                                __HIP_MEMORY_SCOPE_AGENT) != 0u) {
+            got = 1u;
+            break;
+          }
+          __builtin_amdgcn_s_sleep(4);
+        }
+        if (got)
+          __hip_atomic_store(flag, 0u, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_AGENT);  // re-arm
+        s_got = got;
+      }
+      __syncthreads();
+      if (s_got) {
+        for (int i = N_P2 / 2 + tid; i < N_P2; i += 256)
+          p2[i] = gw_ld<true>(p2_ws + (int64_t)b * N_P2 + i);
+      } else {  // sibling never ran: do its channels from our own p1
+        for (int i = N_P2 / 2 + tid; i < N_P2; i += 256) {
+          const int k = i / 16, ph = (i / 4) % 4, pw = i % 4;
+          float q00 = b2[k], q01 = q00, q10 = q00, q11 = q00;
+          #pragma unroll
+          for (int c = 0; c < 10; ++c) {
+            const float* pp = p1 + c * 144 + ph * 2 * 12 + pw * 2;
+            const float* wc = w2 + (k * 10 + c) * 25;  // global: our
+            #pragma unroll                             // w2s half only
+            for (int r = 0; r < 5; ++r) {
+              #pragma unroll
+              for (int s = 0; s < 5; ++s) {
+                const float w = wc[r * 5 + s];
+                const float* pr = pp + r * 12 + s;
+                q00 += pr[0] * w;
+                q01 += pr[1] * w;
+                q10 += pr[12] * w;
+                q11 += pr[13] * w;
+              }
+            }
+          }
+          if (training) {
+            const uint32_t rr = mix32(seed, (uint64_t)b * N_C2K + k);
+            const float dsc = (rr >= 0x80000000u) ? 2.f : 0.f;
+            q00 *= dsc; q01 *= dsc; q10 *= dsc; q11 *= dsc;
+          }
+          float m = fmaxf(fmaxf(q00, q01), fmaxf(q10, q11));
+          p2[i] = m > 0.f ? m : 0.f;
+        }
+      }
+      __syncthreads();
+    }
 
     // fc1 (320->50) + relu + dropout: 4 threads per output, shfl reduce
     if (tid < N_H1 * 4) {
@@ -1040,14 +1153,56 @@ net_fused_fwd_kernel(
   float lsum = 0.f;
   for (int b = blockIdx.x; b < B; b += gridDim.x) {
     const float lp_t = net_fwd_sample(
-        b, tid, B, training, seed, x, w1, b1, w2, b2, wf1, bf1, wf2, bf2,
-        tgt, p1_ws, idx1_ws, m2_ws, p2_ws, idx2_ws, h1_ws, m3_ws, d3_ws,
-        logp_ws, xs, w1s, p1, w2s, p2, d3, logits);
+        b, tid, B, training, seed, 0, nullptr, x, w1, b1, w2, b2, wf1,
+        bf1, wf2, bf2, tgt, p1_ws, idx1_ws, m2_ws, p2_ws, idx2_ws,
+        h1_ws, m3_ws, d3_ws, logp_ws, xs, w1s, p1, w2s, p2, d3, logits);
     if (tid == 0) lsum += -lp_t / B;
   }
   if (tid == 0) {
     if (loss_part) loss_part[blockIdx.x] = lsum;
     else atomicAdd(loss, lsum);
+  }
+}
+
+// Two sibling workgroups per sample (small-B fused fwd: B=128 alone
+// leaves half the 256 CUs idle).  blockIdx.x = 2*b + half; sibling 1
+// computes conv2 channels 10..19, publishes its p2 half (sc1 + flag)
+// and exits; sibling 0 runs the rest.  Grid is exactly 2*B.
+__global__ void
+__launch_bounds__(256)
+net_fused_fwd_split_kernel(
+    const float* __restrict__ x,
+    const float* __restrict__ w1, const float* __restrict__ b1,
+    const float* __restrict__ w2, const float* __restrict__ b2,
+    const float* __restrict__ wf1, const float* __restrict__ bf1,
+    const float* __restrict__ wf2, const float* __restrict__ bf2,
+    const int64_t* __restrict__ tgt,
+    float* __restrict__ p1_ws, uint8_t* __restrict__ idx1_ws,
+    uint8_t* __restrict__ m2_ws, float* __restrict__ p2_ws,
+    uint8_t* __restrict__ idx2_ws, float* __restrict__ h1_ws,
+    uint8_t* __restrict__ m3_ws, float* __restrict__ d3_ws,
+    float* __restrict__ logp_ws, float* __restrict__ loss,
+    float* __restrict__ loss_part,
+    const unsigned long long* __restrict__ seed_p,
+    int B, int training, unsigned int* __restrict__ flags) {
+  __shared__ __attribute__((aligned(16))) float xs[784];
+  __shared__ float w1s[N_C1K * 25 + N_C1K];
+  __shared__ float p1[N_P1];
+  __shared__ float w2s[N_C2K * 10 * 25 + N_C2K];
+  __shared__ float p2[N_P2];
+  __shared__ float d3[N_H1];
+  __shared__ float logits[N_CLS];
+  const int tid = threadIdx.x;
+  const uint64_t seed = seed_p[0];
+  const int b = blockIdx.x >> 1, half = blockIdx.x & 1;
+  const float lp_t = net_fwd_sample<true>(
+      b, tid, B, training, seed, half, flags + b, x, w1, b1, w2, b2,
+      wf1, bf1, wf2, bf2, tgt, p1_ws, idx1_ws, m2_ws, p2_ws, idx2_ws,
+      h1_ws, m3_ws, d3_ws, logp_ws, xs, w1s, p1, w2s, p2, d3, logits);
+  if (tid == 0) {
+    const float lsum = half == 0 ? -lp_t / B : 0.f;
+    if (loss_part) loss_part[blockIdx.x] = lsum;
+    else if (half == 0) atomicAdd(loss, lsum);
   }
 }
 
@@ -1301,9 +1456,9 @@ net_fused_fwdbwd_kernel(
   float lsum = 0.f;
   for (int b = blockIdx.x; b < B; b += gridDim.x) {
     const float lp_t = net_fwd_sample(
-        b, tid, B, training, seed, x, w1, b1, w2, b2, wf1, bf1, wf2, bf2,
-        tgt, p1_ws, idx1_ws, m2_ws, p2_ws, idx2_ws, h1_ws, m3_ws, d3_ws,
-        logp_ws, xs, w1s, p1, w2s, p2, d3, logits);
+        b, tid, B, training, seed, 0, nullptr, x, w1, b1, w2, b2, wf1,
+        bf1, wf2, bf2, tgt, p1_ws, idx1_ws, m2_ws, p2_ws, idx2_ws,
+        h1_ws, m3_ws, d3_ws, logp_ws, xs, w1s, p1, w2s, p2, d3, logits);
     if (tid == 0) lsum += -lp_t / B;
     __syncthreads();  // fwd LDS dead + this block's global stashes visible
     net_bwd_sample(b, 0, 1, tid, B, training, sc, wf1, wf2, tgt,
@@ -1347,30 +1502,6 @@ static_assert(T_CONV2 == N_C2K, "conv2 gw tiling: one tile per output "
 // r), 200 lanes, owns all FIVE sx weights of its (c, r) row: one
 // 12-value xr window + one 8-value gr row feed 40 FMA per channel.
 // Staging is double-buffered float4.
-// partial-row access, parameterized on coherence: the in-launch fold
-// variant stores partials sc1 (device-coherent, straight to the
-// coherence point) and reads them back sc1 — ZERO fences, so the
-// concurrently-running tile blocks keep their L1/L2 activation
-// locality.  The first fold attempt used the plain-store + agent
-// release/acquire recipe and measured 23-28% SLOWER end to end: one
-// `buffer_wbl2` L2 writeback per block x 1744 blocks evicted the very
-// activations the partial blocks re-read (ledger).  Partial traffic is
-// ~1.8 MB/step, so the slower sc1 store path is immaterial.
-template <bool SC1>
-__device__ __forceinline__ void gw_st(float* p, float v) {
-  if (SC1)
-    __hip_atomic_store(p, v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-  else
-    *p = v;
-}
-template <bool SC1>
-__device__ __forceinline__ float gw_ld(const float* p) {
-  if (SC1)
-    return __hip_atomic_load(p, __ATOMIC_RELAXED,
-                             __HIP_MEMORY_SCOPE_AGENT);
-  return *p;
-}
-
 template <int NK, bool SC1 = false>
 __device__ __forceinline__ void net_gw_conv2_fold(
     int k0, int tid, int b0, int b1, float* __restrict__ my,
@@ -2024,9 +2155,9 @@ net_step_kernel(
   float lsum = 0.f;
   for (int b = wg; b < B; b += nblk) {
     const float lp_t = net_fwd_sample(
-        b, tid, B, training, seed, x, w1, b1, w2, b2, wf1, bf1, wf2, bf2,
-        tgt, p1_ws, idx1_ws, m2_ws, p2_ws, idx2_ws, h1_ws, m3_ws, d3_ws,
-        logp_ws, xs, w1s, p1, w2s, p2, d3, logits);
+        b, tid, B, training, seed, 0, nullptr, x, w1, b1, w2, b2, wf1,
+        bf1, wf2, bf2, tgt, p1_ws, idx1_ws, m2_ws, p2_ws, idx2_ws,
+        h1_ws, m3_ws, d3_ws, logp_ws, xs, w1s, p1, w2s, p2, d3, logits);
     if (tid == 0) lsum += -lp_t / B;
   }
   if (tid == 0) loss_part[wg] = lsum;
@@ -2388,6 +2519,45 @@ static unsigned int* gw_cnt_buf(hipStream_t s) {
   return bufs[dev];
 }
 
+// fwd sibling split: 2 workgroups per sample when one per sample
+// cannot fill the 256 CUs.  DTP_FWD_SPLIT=1 forces off, =2 forces on.
+static int fwd_split(int B) {
+  static int env_v = -2;
+  if (env_v == -2) {
+    const char* e = std::getenv("DTP_FWD_SPLIT");
+    env_v = e ? std::atoi(e) : -1;
+  }
+  if (env_v == 1) return 1;
+  if (env_v == 2) return 2;
+  // 3-rep A/B (gpurun_out/fsplit_reps.log): +1.5% at B=64, +0.6% at
+  // B=128, -2.5% at B=192 (the pair grid exceeds the CU count there
+  // and the exchange round trip is pure overhead) — so the split is
+  // a small-batch lever only.
+  return B <= 128 ? 2 : 1;
+}
+
+// fwd grid size (the combine kernels fold loss_part over exactly this
+// many per-block partials)
+static int fwd_grid(int B) {
+  const int sp = fwd_split(B);
+  return sp == 2 ? 2 * B : grid_for(B, 1);
+}
+
+// per-device per-sample handshake flags for the split fwd (zeroed
+// once; consumed-and-re-armed within each launch)
+static unsigned int* fwd_flags_buf(hipStream_t s) {
+  static unsigned int* bufs[64] = {};
+  int dev = 0;
+  HIP_CHECK(hipGetDevice(&dev));
+  if (dev < 0 || dev >= 64) throw std::runtime_error("device index > 64");
+  if (!bufs[dev]) {
+    HIP_CHECK(hipMalloc(&bufs[dev], 4096 * sizeof(unsigned int)));
+    HIP_CHECK(hipMemsetAsync(bufs[dev], 0, 4096 * sizeof(unsigned int),
+                             s));
+  }
+  return bufs[dev];
+}
+
 void net_fused_fwd(uintptr_t x, uintptr_t w1, uintptr_t b1, uintptr_t w2,
                    uintptr_t b2, uintptr_t wf1, uintptr_t bf1,
                    uintptr_t wf2, uintptr_t bf2, uintptr_t tgt,
@@ -2407,6 +2577,21 @@ void net_fused_fwd(uintptr_t x, uintptr_t w1, uintptr_t b1, uintptr_t w2,
     hipLaunchKernelGGL(step_prologue_kernel, dim3(1), dim3(64), 0,
                        S(stream), (unsigned long long*)nullptr,
                        (float*)loss);
+  if (fwd_split(B) == 2) {
+    hipLaunchKernelGGL(net_fused_fwd_split_kernel, dim3(2 * B), dim3(256),
+                       0, S(stream), (const float*)x, (const float*)w1,
+                       (const float*)b1, (const float*)w2,
+                       (const float*)b2, (const float*)wf1,
+                       (const float*)bf1, (const float*)wf2,
+                       (const float*)bf2, (const int64_t*)tgt,
+                       (float*)p1_ws, (uint8_t*)idx1_ws, (uint8_t*)m2_ws,
+                       (float*)p2_ws, (uint8_t*)idx2_ws, (float*)h1_ws,
+                       (uint8_t*)m3_ws, (float*)d3_ws, (float*)logp_ws,
+                       (float*)loss, (float*)loss_part,
+                       (const unsigned long long*)seed_dev, B,
+                       training ? 1 : 0, fwd_flags_buf(S(stream)));
+    return;
+  }
   hipLaunchKernelGGL(net_fused_fwd_kernel, dim3(grid_for(B, 1)), dim3(256),
                      0, S(stream), (const float*)x, (const float*)w1,
                      (const float*)b1, (const float*)w2, (const float*)b2,
@@ -2534,7 +2719,7 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
                        (const float*)glog_ws, (float*)part_ws, B, bchunk,
                        gp, none, none, 0.f, 0.f,
                        (const float*)loss_part, (float*)loss_out,
-                       grid_for(B, 1), sb, gw_cnt_buf(S(stream)));
+                       fwd_grid(B), sb, gw_cnt_buf(S(stream)));
     return;
   }
   hipLaunchKernelGGL(net_gw_partial_kernel, dim3(GW_TILES, nch),
@@ -2550,7 +2735,7 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
                      S(stream), (const float*)part_ws, gp, nch,
                      gw_c1_subs(bchunk) - 1,
                      (const float*)loss_part, (float*)loss_out,
-                     grid_for(B, 1), sb);
+                     fwd_grid(B), sb);
 }
 
 // Combined fwd+bwd (one dispatch) + gw partial + combine: a 3-dispatch
@@ -2695,7 +2880,7 @@ void net_fused_bwd_sgd(uintptr_t x, uintptr_t w2, uintptr_t wf1,
                        (const float*)glog_ws, (float*)part_ws, B, bchunk,
                        gp, pp, bp, (float)lr, (float)mu,
                        (const float*)loss_part, (float*)loss_out,
-                       grid_for(B, 1), sb, gw_cnt_buf(S(stream)));
+                       fwd_grid(B), sb, gw_cnt_buf(S(stream)));
     return;
   }
   hipLaunchKernelGGL(net_gw_partial_kernel, dim3(GW_TILES, nch),
@@ -2711,7 +2896,7 @@ void net_fused_bwd_sgd(uintptr_t x, uintptr_t w2, uintptr_t wf1,
                      S(stream), (const float*)part_ws, gp, pp, bp, nch,
                      gw_c1_subs(bchunk) - 1,
                      (float)lr, (float)mu, (const float*)loss_part,
-                     (float*)loss_out, grid_for(B, 1), sb);
+                     (float*)loss_out, fwd_grid(B), sb);
 }
 
 // raw combine launch (microbenchmarks: time the combine/sgd dispatch
