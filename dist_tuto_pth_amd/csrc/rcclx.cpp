@@ -603,6 +603,16 @@ class TcpMesh {
               continue;
             }
             if (k < 8) continue;  // header still arriving
+            if (hdr != nbytes) continue;
+            // ^ a frame of a DIFFERENT size is not ours: a peer that
+            // already delivered its message may queue traffic for a
+            // later operation on the same pair socket (e.g. its
+            // destroy() barrier token, 4 B) while this any-source
+            // receive still waits on other ranks.  Leave it queued
+            // for the targeted recv that will want it — consuming it
+            // here corrupted the barrier (caught by
+            // tests/test_tcp_backend.py::test_any_source_recv under
+            // repetition).
             read_frame(fds_[p], ptr, nbytes, p);
             return p;
           }

@@ -195,11 +195,32 @@ def _parse_init(init_method: str, world_size: int, rank: int):
 
 
 def _free_port() -> int:
-    s = socket.socket()
-    s.bind(("127.0.0.1", 0))
-    port = s.getsockname()[1]
-    s.close()
-    return port
+    """A port P such that BOTH P and P+1 were bindable at probe time.
+    The native backends put their store on MASTER_PORT+1 (the launcher
+    rendezvous owns MASTER_PORT itself under torchrun), and probing
+    only P let a live ephemeral connection on P+1 fail rank 0's store
+    bind roughly once per ~15 heavy multi-process test runs — the
+    peers then sat in connect timeouts, which looked like a hang in
+    the any-source test.  Still TOCTOU-racy (launch() retries on a
+    fresh pair for that), but the window is now the bind gap, not a
+    whole unprobed port."""
+    for _ in range(64):
+        a = socket.socket()
+        a.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+        a.bind(("127.0.0.1", 0))
+        port = a.getsockname()[1]
+        b = socket.socket()
+        b.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+        try:
+            b.bind(("127.0.0.1", port + 1))
+        except OSError:
+            a.close()
+            b.close()
+            continue
+        b.close()
+        a.close()
+        return port
+    raise RuntimeError("no free adjacent port pair found")
 
 
 # --------------------------------------------------------------------------

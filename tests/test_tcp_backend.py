@@ -132,3 +132,27 @@ def test_no_torch_distributed():
 
 def test_ring_algorithms_on_tcp():
     launch(_fn_ring_algorithms, 3, backend="tcp")
+
+
+def _fn_any_source_skips_foreign_sizes(rank, size):
+    # regression (rcclx.cpp recv_any): an any-source receive must not
+    # consume a queued frame of a DIFFERENT size — that frame belongs
+    # to a later targeted recv (the original bug ate a peer's 4 B
+    # destroy-barrier token mid-wait and corrupted the barrier).
+    import time
+    if rank == 0:
+        t8, t4 = torch.zeros(2), torch.zeros(1)
+        src = dist.recv(t8, src=None)   # rank 1's 4 B frame is already
+        assert src == 2                 # queued; must be skipped
+        assert t8[0].item() == 20.0
+        dist.recv(t4, src=1)
+        assert t4[0].item() == 10.0
+    elif rank == 1:
+        dist.send(torch.full((1,), 10.0), dst=0)
+    else:
+        time.sleep(0.3)
+        dist.send(torch.full((2,), 20.0), dst=0)
+
+
+def test_any_source_skips_foreign_sizes():
+    launch(_fn_any_source_skips_foreign_sizes, 3, backend="tcp")
