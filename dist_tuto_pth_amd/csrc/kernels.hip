@@ -1521,8 +1521,12 @@ __device__ __forceinline__ void net_gw_conv2_fold(
   float be[NK] = {}, bo[NK] = {};
   for (int i = tid; i < NK * 251; i += 256) wacc[i / 251][i % 251] = 0.f;
   // prologue: stage b0 into buffer 0 (sp1 rows and the per-sample
-  // bases are 16 B aligned)
-  {
+  // bases are 16 B aligned).  b0 < b1 guard: ceil-rounded chunk counts
+  // leave EMPTY tail rows for non-divisible B (e.g. B=100, nch=16,
+  // bchunk=7 -> row 15 starts at sample 105) — staging unconditionally
+  // read up to ~6 KB past p1_ws/ga2_ws for those rows; the zeroed wacc
+  // flush below is all an empty row needs.
+  if (b0 < b1) {
     float4* d4 = reinterpret_cast<float4*>(sp1[0]);
     const float4* s4 = reinterpret_cast<const float4*>(
         p1_ws + (int64_t)b0 * N_P1);
@@ -1761,11 +1765,18 @@ net_gw_partial_kernel(int c1_subs, const float* __restrict__ x,
                       const float* __restrict__ ga2_ws,
                       const float* __restrict__ gh1_ws,
                       const float* __restrict__ glog_ws,
-                      float* __restrict__ part,  // [nch][GW_ROW]
-                      int B, int bchunk, int tile_base) {
-  const int b0 = blockIdx.y * bchunk;
-  net_gw_tile(blockIdx.x + tile_base, threadIdx.x, b0,
-              min(B, b0 + bchunk), c1_subs,
+                      float* __restrict__ part,  // [max(nch,nch2)][GW_ROW]
+                      int B, int bchunk, int bchunk2, int nch, int nch2,
+                      int tile_base) {
+  // conv2 runs at its own chunk count nch2 (it is the straggler family
+  // at small B); gridDim.y = max(nch, nch2), families idle on the rows
+  // beyond their own count (idle blocks are launch-granularity free)
+  const int tile = blockIdx.x + tile_base;
+  const bool c2 = tile < T_CONV2;
+  if ((int)blockIdx.y >= (c2 ? nch2 : nch)) return;
+  const int bc = c2 ? bchunk2 : bchunk;
+  const int b0 = blockIdx.y * bc;
+  net_gw_tile(tile, threadIdx.x, b0, min(B, b0 + bc), c1_subs,
               part + (int64_t)blockIdx.y * GW_ROW,
               x, p1_ws, p2_ws, d3_ws, ga1_ws, ga2_ws, gh1_ws, glog_ws);
 }
@@ -1776,18 +1787,25 @@ struct GwPtrs { float* p[8]; };
 // sum flat-grad element i over the nch chunk rows (+ conv1 extension)
 template <bool SC1 = false>
 __device__ __forceinline__ float net_gw_combine_elem(
-    int i, int nch, int c1_ext, const float* __restrict__ part) {
+    int i, int nch, int nch2, int c1_ext,
+    const float* __restrict__ part) {
   // four independent accumulator chains: the single 32-deep
-  // load+add chain was latency-bound (VALUBusy ~0, profiles/)
+  // load+add chain was latency-bound (VALUBusy ~0, profiles/).
+  // nch2: the conv2 family may run at its OWN chunk count (its tile
+  // columns were the gw straggler at small B; more chunks = more
+  // parallelism there without paying the conv1-extension fold cost a
+  // global chunk increase incurs) — conv2-region elements fold nch2
+  // rows, everything else nch.
+  const int n = (i >= OFF_W2 && i < OFF_WF1) ? nch2 : nch;
   float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
   int c = 0;
-  for (; c + 3 < nch; c += 4) {
+  for (; c + 3 < n; c += 4) {
     a0 += gw_ld<SC1>(part + (int64_t)c * GW_ROW + i);
     a1 += gw_ld<SC1>(part + (int64_t)(c + 1) * GW_ROW + i);
     a2 += gw_ld<SC1>(part + (int64_t)(c + 2) * GW_ROW + i);
     a3 += gw_ld<SC1>(part + (int64_t)(c + 3) * GW_ROW + i);
   }
-  for (; c < nch; ++c) a0 += gw_ld<SC1>(part + (int64_t)c * GW_ROW + i);
+  for (; c < n; ++c) a0 += gw_ld<SC1>(part + (int64_t)c * GW_ROW + i);
   if (i < 260) {  // conv1 sub-block extension rows (see GW_ROW)
     // c1_ext is 7 (band mode) or 23 (row mode): branch to fully
     // unrolled folds — a runtime-bound loop here cost the combine
@@ -1824,11 +1842,13 @@ __device__ __forceinline__ float net_gw_combine_elem(
 // ran a latency-exposed 8-deep L2 chain (profiles/).  Returns the full
 // sum on EVERY lane of the quad (butterfly reduction).
 __device__ __forceinline__ float net_gw_combine_elem_quad(
-    int i, int q, int nch, int c1_ext, const float* __restrict__ part) {
+    int i, int q, int nch, int nch2, int c1_ext,
+    const float* __restrict__ part) {
+  const int n = (i >= OFF_W2 && i < OFF_WF1) ? nch2 : nch;
   float a0 = 0.f, a1 = 0.f;
-  for (int c = q; c < nch; c += 8)
+  for (int c = q; c < n; c += 8)
     a0 += part[(int64_t)c * GW_ROW + i];
-  for (int c = q + 4; c < nch; c += 8)
+  for (int c = q + 4; c < n; c += 8)
     a1 += part[(int64_t)c * GW_ROW + i];
   if (i < 260) {  // conv1 extension rows, spread across the quad
     // fully unrolled per mode (see net_gw_combine_elem note)
@@ -1891,7 +1911,8 @@ __device__ __forceinline__ void net_loss_finalize(
 }
 
 __global__ void net_gw_combine_kernel(const float* __restrict__ part,
-                                      GwPtrs g, int nch, int c1_ext,
+                                      GwPtrs g, int nch, int nch2,
+                                      int c1_ext,
                                       const float* __restrict__ loss_part,
                                       float* __restrict__ loss_out,
                                       int nblk_fwd,
@@ -1902,7 +1923,8 @@ __global__ void net_gw_combine_kernel(const float* __restrict__ part,
        t4 < (int64_t)GW_TOTAL * 4;
        t4 += (int64_t)gridDim.x * blockDim.x) {
     const int i = (int)(t4 >> 2), q = (int)(t4 & 3);
-    const float acc = net_gw_combine_elem_quad(i, q, nch, c1_ext, part);
+    const float acc = net_gw_combine_elem_quad(i, q, nch, nch2, c1_ext,
+                                               part);
     if (q == 0) {
       const int t = net_gw_tensor_of(i, off);
       g.p[t][i - off[t]] = acc;
@@ -1919,7 +1941,8 @@ __global__ void net_gw_combine_kernel(const float* __restrict__ part,
 // the momentum buffer and the parameter exactly like sgd_step_kernel.
 __global__ void net_gw_combine_sgd_kernel(const float* __restrict__ part,
                                           GwPtrs g, GwPtrs prm,
-                                          GwPtrs buf, int nch, int c1_ext,
+                                          GwPtrs buf, int nch, int nch2,
+                                          int c1_ext,
                                           float lr,
                                           float mu,
                                           const float* __restrict__ loss_part,
@@ -1932,7 +1955,8 @@ __global__ void net_gw_combine_sgd_kernel(const float* __restrict__ part,
        t4 < (int64_t)GW_TOTAL * 4;
        t4 += (int64_t)gridDim.x * blockDim.x) {
     const int i = (int)(t4 >> 2), q = (int)(t4 & 3);
-    const float acc = net_gw_combine_elem_quad(i, q, nch, c1_ext, part);
+    const float acc = net_gw_combine_elem_quad(i, q, nch, nch2, c1_ext,
+                                               part);
     if (q == 0) {
       const int t = net_gw_tensor_of(i, off);
       const int64_t j = i - off[t];
@@ -1977,7 +2001,8 @@ __device__ __forceinline__ void net_gw_fold_range(
   const int off[9] = {OFF_W1, OFF_B1, OFF_W2, OFF_B2, OFF_WF1, OFF_BF1,
                       OFF_WF2, OFF_BF2, GW_TOTAL};
   for (int i = i0 + (int)threadIdx.x; i < i1; i += 256) {
-    const float acc = net_gw_combine_elem<true>(i, nch, c1_ext, part);
+    const float acc = net_gw_combine_elem<true>(i, nch, nch, c1_ext,
+                                                part);
     const int t = net_gw_tensor_of(i, off);
     const int64_t j = i - off[t];
     g.p[t][j] = acc;
@@ -2189,7 +2214,7 @@ net_step_kernel(
   const int off[9] = {OFF_W1, OFF_B1, OFF_W2, OFF_B2, OFF_WF1, OFF_BF1,
                       OFF_WF2, OFF_BF2, GW_TOTAL};
   for (int i = wg * 256 + tid; i < GW_TOTAL; i += nblk * 256) {
-    const float acc = net_gw_combine_elem(i, nch, 7, part);
+    const float acc = net_gw_combine_elem(i, nch, nch, 7, part);
     const int t = net_gw_tensor_of(i, off);
     const int64_t j = i - off[t];
     grd.p[t][j] = acc;
@@ -2558,6 +2583,26 @@ static unsigned int* fwd_flags_buf(hipStream_t s) {
   return bufs[dev];
 }
 
+// conv2-family gw chunk count (DTP_GW_NCH2 overrides).  The conv2
+// pair-mode columns are the gw straggler at small B (20.2 us isolated
+// of the 29.5 us bundle at B=128, kernel_micro); running ONLY conv2 at
+// a higher chunk count buys its parallelism without the conv1
+// extension-fold cost that made a GLOBAL chunk increase lose.
+static int gw_nch2(int B, int nch) {
+  static int env_nch = -2;
+  if (env_nch == -2) {
+    const char* e = std::getenv("DTP_GW_NCH2");
+    env_nch = e ? std::atoi(e) : -1;
+  }
+  // sweep (gpurun_out/nch2_sweep.log): B=128 nch2=24 74.6us vs 76.2 at
+  // the uniform 16 (+2.1%); 32 within noise of 24; B=512 nch2=32 LOSES
+  // (combine cost) — so only the small-batch band deviates from nch.
+  int nch2 = env_nch > 0 ? env_nch : (B <= 192 ? 24 : nch);
+  if (nch2 > 32) nch2 = 32;  // part workspace holds 32 rows (_ws)
+  if (nch2 > B) nch2 = B;
+  return nch2;
+}
+
 void net_fused_fwd(uintptr_t x, uintptr_t w1, uintptr_t b1, uintptr_t w2,
                    uintptr_t b2, uintptr_t wf1, uintptr_t bf1,
                    uintptr_t wf2, uintptr_t bf2, uintptr_t tgt,
@@ -2701,6 +2746,10 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
   // combine, more batch per partial block — sweep on hardware).
   const int nch = gw_nch(B);
   const int bchunk = (B + nch - 1) / nch;
+  // fold path keeps uniform chunks (its per-column tickets assume it)
+  const int nch2 = gw_fold_on() ? nch : gw_nch2(B, nch);
+  const int bchunk2 = (B + nch2 - 1) / nch2;
+  const int gw_gy = nch > nch2 ? nch : nch2;
   GwPtrs gp;
   gp.p[0] = (float*)gw1; gp.p[1] = (float*)gb1;
   gp.p[2] = (float*)gw2; gp.p[3] = (float*)gb2;
@@ -2722,17 +2771,17 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
                        fwd_grid(B), sb, gw_cnt_buf(S(stream)));
     return;
   }
-  hipLaunchKernelGGL(net_gw_partial_kernel, dim3(GW_TILES, nch),
+  hipLaunchKernelGGL(net_gw_partial_kernel, dim3(GW_TILES, gw_gy),
                      dim3(256), 0, S(stream), gw_c1_subs(bchunk),
                      (const float*)x,
                      (const float*)p1_ws, (const float*)p2_ws,
                      (const float*)d3_ws, (const float*)ga1_ws,
                      (const float*)ga2_ws, (const float*)gh1_ws,
                      (const float*)glog_ws, (float*)part_ws, B, bchunk,
-                     0);
+                     bchunk2, nch, nch2, 0);
   hipLaunchKernelGGL(net_gw_combine_kernel,
                      dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
-                     S(stream), (const float*)part_ws, gp, nch,
+                     S(stream), (const float*)part_ws, gp, nch, nch2,
                      gw_c1_subs(bchunk) - 1,
                      (const float*)loss_part, (float*)loss_out,
                      fwd_grid(B), sb);
@@ -2774,6 +2823,10 @@ void net_fused_fwdbwd(
                      training ? 1 : 0);
   const int nch = gw_nch(B);
   const int bchunk = (B + nch - 1) / nch;
+  // fold path keeps uniform chunks (its per-column tickets assume it)
+  const int nch2 = gw_fold_on() ? nch : gw_nch2(B, nch);
+  const int bchunk2 = (B + nch2 - 1) / nch2;
+  const int gw_gy = nch > nch2 ? nch : nch2;
   GwPtrs gp{}, pp{}, bp{};
   for (int i = 0; i < 8; ++i) gp.p[i] = (float*)grd_v[i];
   const bool sgd = !prm_v.empty();
@@ -2799,26 +2852,26 @@ void net_fused_fwdbwd(
                        sb, gw_cnt_buf(S(stream)));
     return;
   }
-  hipLaunchKernelGGL(net_gw_partial_kernel, dim3(GW_TILES, nch),
+  hipLaunchKernelGGL(net_gw_partial_kernel, dim3(GW_TILES, gw_gy),
                      dim3(256), 0, S(stream), gw_c1_subs(bchunk),
                      (const float*)x,
                      (const float*)p1_ws, (const float*)p2_ws,
                      (const float*)d3_ws, (const float*)ga1_ws,
                      (const float*)ga2_ws, (const float*)gh1_ws,
                      (const float*)glog_ws, (float*)part_ws, B, bchunk,
-                     0);
+                     bchunk2, nch, nch2, 0);
   if (!sgd) {
     hipLaunchKernelGGL(net_gw_combine_kernel,
                        dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
                        S(stream), (const float*)part_ws, gp, nch,
-                       gw_c1_subs(bchunk) - 1,
+                       nch2, gw_c1_subs(bchunk) - 1,
                        (const float*)loss_part, (float*)loss_out,
                        nblk, sb);
   } else {
     hipLaunchKernelGGL(net_gw_combine_sgd_kernel,
                        dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
-                       S(stream), (const float*)part_ws, gp, pp, bp, nch,
-                       gw_c1_subs(bchunk) - 1,
+                       S(stream), (const float*)part_ws, gp, pp, bp,
+                       nch, nch2, gw_c1_subs(bchunk) - 1,
                        (float)lr, (float)mu, (const float*)loss_part,
                        (float*)loss_out, nblk, sb);
   }
@@ -2862,6 +2915,10 @@ void net_fused_bwd_sgd(uintptr_t x, uintptr_t w2, uintptr_t wf1,
                      training ? 1 : 0, split);
   const int nch = gw_nch(B);
   const int bchunk = (B + nch - 1) / nch;
+  // fold path keeps uniform chunks (its per-column tickets assume it)
+  const int nch2 = gw_fold_on() ? nch : gw_nch2(B, nch);
+  const int bchunk2 = (B + nch2 - 1) / nch2;
+  const int gw_gy = nch > nch2 ? nch : nch2;
   GwPtrs gp{}, pp{}, bp{};
   for (int i = 0; i < 8; ++i) {
     gp.p[i] = (float*)grd_v[i];
@@ -2883,18 +2940,18 @@ void net_fused_bwd_sgd(uintptr_t x, uintptr_t w2, uintptr_t wf1,
                        fwd_grid(B), sb, gw_cnt_buf(S(stream)));
     return;
   }
-  hipLaunchKernelGGL(net_gw_partial_kernel, dim3(GW_TILES, nch),
+  hipLaunchKernelGGL(net_gw_partial_kernel, dim3(GW_TILES, gw_gy),
                      dim3(256), 0, S(stream), gw_c1_subs(bchunk),
                      (const float*)x,
                      (const float*)p1_ws, (const float*)p2_ws,
                      (const float*)d3_ws, (const float*)ga1_ws,
                      (const float*)ga2_ws, (const float*)gh1_ws,
                      (const float*)glog_ws, (float*)part_ws, B, bchunk,
-                     0);
+                     bchunk2, nch, nch2, 0);
   hipLaunchKernelGGL(net_gw_combine_sgd_kernel,
                      dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
-                     S(stream), (const float*)part_ws, gp, pp, bp, nch,
-                     gw_c1_subs(bchunk) - 1,
+                     S(stream), (const float*)part_ws, gp, pp, bp,
+                     nch, nch2, gw_c1_subs(bchunk) - 1,
                      (float)lr, (float)mu, (const float*)loss_part,
                      (float*)loss_out, fwd_grid(B), sb);
 }
@@ -2903,13 +2960,13 @@ void net_fused_bwd_sgd(uintptr_t x, uintptr_t w2, uintptr_t wf1,
 // in isolation)
 void net_gw_combine_raw(uintptr_t part_ws,
                         const std::vector<uintptr_t>& grd_v, int nch,
-                        int c1_ext, uintptr_t stream) {
+                        int nch2, int c1_ext, uintptr_t stream) {
   GwPtrs gp{};
   for (int i = 0; i < 8; ++i) gp.p[i] = (float*)grd_v[i];
   hipLaunchKernelGGL(net_gw_combine_kernel,
                      dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
-                     S(stream), (const float*)part_ws, gp, nch, c1_ext,
-                     nullptr, nullptr, 0, nullptr);
+                     S(stream), (const float*)part_ws, gp, nch, nch2,
+                     c1_ext, nullptr, nullptr, 0, nullptr);
 }
 
 // raw combine+sgd launch (microbenchmarks)
@@ -2917,7 +2974,8 @@ void net_gw_combine_sgd_raw(uintptr_t part_ws,
                             const std::vector<uintptr_t>& grd_v,
                             const std::vector<uintptr_t>& prm_v,
                             const std::vector<uintptr_t>& buf_v,
-                            int nch, int c1_ext, double lr, double mu,
+                            int nch, int nch2, int c1_ext, double lr,
+                            double mu,
                             uintptr_t loss_part, uintptr_t loss_out,
                             int nblk_fwd, uintptr_t stream) {
   GwPtrs gp{}, pp{}, bp{};
@@ -2929,7 +2987,7 @@ void net_gw_combine_sgd_raw(uintptr_t part_ws,
   hipLaunchKernelGGL(net_gw_combine_sgd_kernel,
                      dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
                      S(stream), (const float*)part_ws, gp, pp, bp, nch,
-                     c1_ext, (float)lr, (float)mu,
+                     nch2, c1_ext, (float)lr, (float)mu,
                      (const float*)loss_part, (float*)loss_out,
                      nblk_fwd, nullptr);
 }
@@ -2940,15 +2998,17 @@ void net_gw_partial_raw(uintptr_t x, uintptr_t p1_ws, uintptr_t p2_ws,
                         uintptr_t d3_ws, uintptr_t ga1_ws,
                         uintptr_t ga2_ws, uintptr_t gh1_ws,
                         uintptr_t glog_ws, uintptr_t part_ws, int B,
-                        int bchunk, int tile_base, int ntiles, int nch,
+                        int bchunk, int bchunk2, int tile_base,
+                        int ntiles, int nch, int nch2,
                         uintptr_t stream) {
-  hipLaunchKernelGGL(net_gw_partial_kernel, dim3(ntiles, nch), dim3(256),
+  const int gy = nch > nch2 ? nch : nch2;
+  hipLaunchKernelGGL(net_gw_partial_kernel, dim3(ntiles, gy), dim3(256),
                      0, S(stream), gw_c1_subs(bchunk), (const float*)x,
                      (const float*)p1_ws, (const float*)p2_ws,
                      (const float*)d3_ws, (const float*)ga1_ws,
                      (const float*)ga2_ws, (const float*)gh1_ws,
                      (const float*)glog_ws, (float*)part_ws, B, bchunk,
-                     tile_base);
+                     bchunk2, nch, nch2, tile_base);
 }
 
 // cooperative grid-barrier cost probe: `nsync` grid.sync()s and
