@@ -340,3 +340,19 @@ def test_fold_optin_matches_default_subprocess():
         g = torch.tensor(got[n]).view_as(w)
         assert torch.allclose(g, w, atol=1e-6), \
             (n, (g - w).abs().max().item())
+
+
+@pytest.mark.parametrize("B", [33, 100, 257])
+def test_fused_step_odd_batch_sizes(B):
+    """Non-divisible batch sizes produce EMPTY tail chunk rows in the
+    gw partial grid (ceil rounding) and exercise the family-specific
+    conv2 chunk count — regression for the unguarded staging prologue
+    that read past the workspaces for such rows."""
+    net_c, net_g, x, tgt = _mk(20 + B, B=B)
+    loss_c = F.nll_loss(net_c(x), tgt)
+    loss_c.backward()
+    net_fused_step(net_g, x.to(DEV), tgt.to(DEV))
+    torch.cuda.synchronize()
+    for (n, pc), pg in zip(net_c.named_parameters(), net_g.parameters()):
+        assert torch.allclose(pg.grad.cpu(), pc.grad, atol=5e-4), \
+            (n, (pg.grad.cpu() - pc.grad).abs().max())
