@@ -95,7 +95,10 @@ def main():
     nch = int(os.environ.get("DTP_GW_NCH", 0)) or         (16 if B <= 192 else (24 if B <= 768 else 32))
     nch = min(nch, 32, B)
     bchunk = (B + nch - 1) // nch
-    nch2 = min(int(os.environ.get("DTP_GW_NCH2", 0)) or nch, 32, B)
+    nch1 = min(int(os.environ.get("DTP_GW_NCH1", 0)) or nch, 32, B)
+    bchunk1 = (B + nch1 - 1) // nch1
+    nch2 = min(int(os.environ.get("DTP_GW_NCH2", 0)) or
+               (24 if B <= 192 else nch), 32, B)
     bchunk2 = (B + nch2 - 1) // nch2
     c1_ext = (8 if bchunk <= 16 else 24) - 1
 
@@ -106,7 +109,7 @@ def main():
                 ws["d3"].data_ptr(), ws["ga1"].data_ptr(),
                 ws["ga2"].data_ptr(), ws["gh1"].data_ptr(),
                 ws["glog"].data_ptr(), ws["part"].data_ptr(), B, bchunk,
-                bchunk2, base, ntiles, nch, nch2, s)
+                bchunk1, bchunk2, base, ntiles, nch, nch1, nch2, s)
         return f
 
     results = {}
@@ -123,13 +126,13 @@ def main():
     results["gw all tiles"] = time_fn(seg(0, T_CONV2 + T_FC1 + T_CONV1 + T_FC2))
     results["gw combine"] = time_fn(lambda: k.net_gw_combine_raw(
         ws["part"].data_ptr(), [p.grad.data_ptr() for p in params], nch,
-        nch2, c1_ext, s))
+        nch1, nch2, c1_ext, s))
     ws.setdefault("loss_part", torch.empty(4096, device=dev))
     bufs = [b.data_ptr() for b in opt._bufs]
     results["gw combine+sgd+loss"] = time_fn(
         lambda: k.net_gw_combine_sgd_raw(
             ws["part"].data_ptr(), [p.grad.data_ptr() for p in params],
-            pp, bufs, nch, nch2, c1_ext, 0.01, 0.5,
+            pp, bufs, nch, nch1, nch2, c1_ext, 0.01, 0.5,
             ws["loss_part"].data_ptr(), ws["loss"].data_ptr(), 128, s))
     results["sgd"] = time_fn(opt.step)
 

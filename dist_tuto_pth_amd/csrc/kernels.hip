@@ -1765,18 +1765,20 @@ net_gw_partial_kernel(int c1_subs, const float* __restrict__ x,
                       const float* __restrict__ ga2_ws,
                       const float* __restrict__ gh1_ws,
                       const float* __restrict__ glog_ws,
-                      float* __restrict__ part,  // [max(nch,nch2)][GW_ROW]
-                      int B, int bchunk, int bchunk2, int nch, int nch2,
-                      int tile_base) {
-  // conv2 runs at its own chunk count nch2 (it is the straggler family
-  // at small B); gridDim.y = max(nch, nch2), families idle on the rows
-  // beyond their own count (idle blocks are launch-granularity free)
+                      float* __restrict__ part,  // [max chunk][GW_ROW]
+                      int B, int bchunk, int bchunk1, int bchunk2,
+                      int nch, int nch1, int nch2, int tile_base) {
+  // conv1/conv2 run at their own chunk counts (they are the straggler
+  // families at small B); gridDim.y = max of the three, families idle
+  // on the rows beyond their own count (idle blocks are
+  // launch-granularity free)
   const int tile = blockIdx.x + tile_base;
   const bool c2 = tile < T_CONV2;
-  if ((int)blockIdx.y >= (c2 ? nch2 : nch)) return;
-  const int bc = c2 ? bchunk2 : bchunk;
-  const int b0 = blockIdx.y * bc;
-  net_gw_tile(tile, threadIdx.x, b0, min(B, b0 + bc), c1_subs,
+  const bool c1 = tile >= T_CONV2 && tile < T_CONV2 + T_CONV1;
+  if ((int)blockIdx.y >= (c2 ? nch2 : c1 ? nch1 : nch)) return;
+  const int bc = c2 ? bchunk2 : c1 ? bchunk1 : bchunk;
+  net_gw_tile(tile, threadIdx.x, (int)blockIdx.y * bc,
+              min(B, ((int)blockIdx.y + 1) * bc), c1_subs,
               part + (int64_t)blockIdx.y * GW_ROW,
               x, p1_ws, p2_ws, d3_ws, ga1_ws, ga2_ws, gh1_ws, glog_ws);
 }
@@ -1787,16 +1789,18 @@ struct GwPtrs { float* p[8]; };
 // sum flat-grad element i over the nch chunk rows (+ conv1 extension)
 template <bool SC1 = false>
 __device__ __forceinline__ float net_gw_combine_elem(
-    int i, int nch, int nch2, int c1_ext,
+    int i, int nch, int nch1, int nch2, int c1_ext,
     const float* __restrict__ part) {
   // four independent accumulator chains: the single 32-deep
   // load+add chain was latency-bound (VALUBusy ~0, profiles/).
-  // nch2: the conv2 family may run at its OWN chunk count (its tile
-  // columns were the gw straggler at small B; more chunks = more
-  // parallelism there without paying the conv1-extension fold cost a
-  // global chunk increase incurs) — conv2-region elements fold nch2
-  // rows, everything else nch.
-  const int n = (i >= OFF_W2 && i < OFF_WF1) ? nch2 : nch;
+  // nch1/nch2: the conv1/conv2 families may run at their OWN chunk
+  // counts (they are the gw stragglers at small B; more chunks = more
+  // parallelism there without paying their fold cost on every other
+  // family, which is what made a GLOBAL chunk increase lose) —
+  // conv1-region elements (i < 260, extensions included) fold nch1
+  // rows, conv2-region elements nch2 rows, everything else nch.
+  const int n = i < 260 ? nch1
+              : (i >= OFF_W2 && i < OFF_WF1) ? nch2 : nch;
   float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
   int c = 0;
   for (; c + 3 < n; c += 4) {
@@ -1811,7 +1815,7 @@ __device__ __forceinline__ float net_gw_combine_elem(
     // unrolled folds — a runtime-bound loop here cost the combine
     // 50% (8.1 -> 12.2 us, r2 ledger)
     if (c1_ext == 7) {
-      for (int c2 = 0; c2 < nch; ++c2) {
+      for (int c2 = 0; c2 < nch1; ++c2) {
         const float* ext = part + (int64_t)c2 * GW_ROW + GW_TOTAL;
         a0 += gw_ld<SC1>(ext + i) + gw_ld<SC1>(ext + 4 * 260 + i);
         a1 += gw_ld<SC1>(ext + 260 + i) + gw_ld<SC1>(ext + 5 * 260 + i);
@@ -1820,7 +1824,7 @@ __device__ __forceinline__ float net_gw_combine_elem(
         a3 += gw_ld<SC1>(ext + 3 * 260 + i);
       }
     } else {
-      for (int c2 = 0; c2 < nch; ++c2) {
+      for (int c2 = 0; c2 < nch1; ++c2) {
         const float* ext = part + (int64_t)c2 * GW_ROW + GW_TOTAL;
         #pragma unroll
         for (int s = 0; s < 23; s += 4) {
@@ -1842,9 +1846,10 @@ __device__ __forceinline__ float net_gw_combine_elem(
 // ran a latency-exposed 8-deep L2 chain (profiles/).  Returns the full
 // sum on EVERY lane of the quad (butterfly reduction).
 __device__ __forceinline__ float net_gw_combine_elem_quad(
-    int i, int q, int nch, int nch2, int c1_ext,
+    int i, int q, int nch, int nch1, int nch2, int c1_ext,
     const float* __restrict__ part) {
-  const int n = (i >= OFF_W2 && i < OFF_WF1) ? nch2 : nch;
+  const int n = i < 260 ? nch1
+              : (i >= OFF_W2 && i < OFF_WF1) ? nch2 : nch;
   float a0 = 0.f, a1 = 0.f;
   for (int c = q; c < n; c += 8)
     a0 += part[(int64_t)c * GW_ROW + i];
@@ -1854,7 +1859,7 @@ __device__ __forceinline__ float net_gw_combine_elem_quad(
     // fully unrolled per mode (see net_gw_combine_elem note)
     if (c1_ext == 7) {
       if (q == 0) {
-        for (int c2 = 0; c2 < nch; ++c2) {
+        for (int c2 = 0; c2 < nch1; ++c2) {
           const float* ext = part + (int64_t)c2 * GW_ROW + GW_TOTAL;
           a0 += ext[i] + ext[260 + i] + ext[2 * 260 + i] +
                 ext[3 * 260 + i];
@@ -1862,7 +1867,7 @@ __device__ __forceinline__ float net_gw_combine_elem_quad(
         }
       }
     } else {
-      for (int c2 = 0; c2 < nch; ++c2) {
+      for (int c2 = 0; c2 < nch1; ++c2) {
         const float* ext = part + (int64_t)c2 * GW_ROW + GW_TOTAL;
         #pragma unroll
         for (int s = 0; s < 23; s += 8) {
@@ -1911,8 +1916,8 @@ __device__ __forceinline__ void net_loss_finalize(
 }
 
 __global__ void net_gw_combine_kernel(const float* __restrict__ part,
-                                      GwPtrs g, int nch, int nch2,
-                                      int c1_ext,
+                                      GwPtrs g, int nch, int nch1,
+                                      int nch2, int c1_ext,
                                       const float* __restrict__ loss_part,
                                       float* __restrict__ loss_out,
                                       int nblk_fwd,
@@ -1923,8 +1928,8 @@ __global__ void net_gw_combine_kernel(const float* __restrict__ part,
        t4 < (int64_t)GW_TOTAL * 4;
        t4 += (int64_t)gridDim.x * blockDim.x) {
     const int i = (int)(t4 >> 2), q = (int)(t4 & 3);
-    const float acc = net_gw_combine_elem_quad(i, q, nch, nch2, c1_ext,
-                                               part);
+    const float acc = net_gw_combine_elem_quad(i, q, nch, nch1, nch2,
+                                               c1_ext, part);
     if (q == 0) {
       const int t = net_gw_tensor_of(i, off);
       g.p[t][i - off[t]] = acc;
@@ -1941,8 +1946,8 @@ __global__ void net_gw_combine_kernel(const float* __restrict__ part,
 // the momentum buffer and the parameter exactly like sgd_step_kernel.
 __global__ void net_gw_combine_sgd_kernel(const float* __restrict__ part,
                                           GwPtrs g, GwPtrs prm,
-                                          GwPtrs buf, int nch, int nch2,
-                                          int c1_ext,
+                                          GwPtrs buf, int nch, int nch1,
+                                          int nch2, int c1_ext,
                                           float lr,
                                           float mu,
                                           const float* __restrict__ loss_part,
@@ -1955,8 +1960,8 @@ __global__ void net_gw_combine_sgd_kernel(const float* __restrict__ part,
        t4 < (int64_t)GW_TOTAL * 4;
        t4 += (int64_t)gridDim.x * blockDim.x) {
     const int i = (int)(t4 >> 2), q = (int)(t4 & 3);
-    const float acc = net_gw_combine_elem_quad(i, q, nch, nch2, c1_ext,
-                                               part);
+    const float acc = net_gw_combine_elem_quad(i, q, nch, nch1, nch2,
+                                               c1_ext, part);
     if (q == 0) {
       const int t = net_gw_tensor_of(i, off);
       const int64_t j = i - off[t];
@@ -2001,8 +2006,8 @@ __device__ __forceinline__ void net_gw_fold_range(
   const int off[9] = {OFF_W1, OFF_B1, OFF_W2, OFF_B2, OFF_WF1, OFF_BF1,
                       OFF_WF2, OFF_BF2, GW_TOTAL};
   for (int i = i0 + (int)threadIdx.x; i < i1; i += 256) {
-    const float acc = net_gw_combine_elem<true>(i, nch, nch, c1_ext,
-                                                part);
+    const float acc = net_gw_combine_elem<true>(i, nch, nch, nch,
+                                                c1_ext, part);
     const int t = net_gw_tensor_of(i, off);
     const int64_t j = i - off[t];
     g.p[t][j] = acc;
@@ -2214,7 +2219,7 @@ net_step_kernel(
   const int off[9] = {OFF_W1, OFF_B1, OFF_W2, OFF_B2, OFF_WF1, OFF_BF1,
                       OFF_WF2, OFF_BF2, GW_TOTAL};
   for (int i = wg * 256 + tid; i < GW_TOTAL; i += nblk * 256) {
-    const float acc = net_gw_combine_elem(i, nch, nch, 7, part);
+    const float acc = net_gw_combine_elem(i, nch, nch, nch, 7, part);
     const int t = net_gw_tensor_of(i, off);
     const int64_t j = i - off[t];
     grd.p[t][j] = acc;
@@ -2588,6 +2593,22 @@ static unsigned int* fwd_flags_buf(hipStream_t s) {
 // of the 29.5 us bundle at B=128, kernel_micro); running ONLY conv2 at
 // a higher chunk count buys its parallelism without the conv1
 // extension-fold cost that made a GLOBAL chunk increase lose.
+// conv1-family gw chunk count (DTP_GW_NCH1 overrides); 16.6 us
+// isolated at B=128 — the second straggler after conv2.  Its combine
+// cost scales with nch1 * c1_ext (extension rows), which is why the
+// global sweep rejected 32; family-local it can still pay.
+static int gw_nch1(int B, int nch) {
+  static int env_nch = -2;
+  if (env_nch == -2) {
+    const char* e = std::getenv("DTP_GW_NCH1");
+    env_nch = e ? std::atoi(e) : -1;
+  }
+  int nch1 = env_nch > 0 ? env_nch : nch;
+  if (nch1 > 32) nch1 = 32;
+  if (nch1 > B) nch1 = B;
+  return nch1;
+}
+
 static int gw_nch2(int B, int nch) {
   static int env_nch = -2;
   if (env_nch == -2) {
@@ -2747,9 +2768,11 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
   const int nch = gw_nch(B);
   const int bchunk = (B + nch - 1) / nch;
   // fold path keeps uniform chunks (its per-column tickets assume it)
+  const int nch1 = gw_fold_on() ? nch : gw_nch1(B, nch);
   const int nch2 = gw_fold_on() ? nch : gw_nch2(B, nch);
+  const int bchunk1 = (B + nch1 - 1) / nch1;
   const int bchunk2 = (B + nch2 - 1) / nch2;
-  const int gw_gy = nch > nch2 ? nch : nch2;
+  const int gw_gy = std::max(nch, std::max(nch1, nch2));
   GwPtrs gp;
   gp.p[0] = (float*)gw1; gp.p[1] = (float*)gb1;
   gp.p[2] = (float*)gw2; gp.p[3] = (float*)gb2;
@@ -2778,11 +2801,11 @@ void net_fused_bwd(uintptr_t x, uintptr_t w2, uintptr_t wf1, uintptr_t wf2,
                      (const float*)d3_ws, (const float*)ga1_ws,
                      (const float*)ga2_ws, (const float*)gh1_ws,
                      (const float*)glog_ws, (float*)part_ws, B, bchunk,
-                     bchunk2, nch, nch2, 0);
+                     bchunk1, bchunk2, nch, nch1, nch2, 0);
   hipLaunchKernelGGL(net_gw_combine_kernel,
                      dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
-                     S(stream), (const float*)part_ws, gp, nch, nch2,
-                     gw_c1_subs(bchunk) - 1,
+                     S(stream), (const float*)part_ws, gp, nch, nch1,
+                     nch2, gw_c1_subs(bchunk) - 1,
                      (const float*)loss_part, (float*)loss_out,
                      fwd_grid(B), sb);
 }
@@ -2824,9 +2847,11 @@ void net_fused_fwdbwd(
   const int nch = gw_nch(B);
   const int bchunk = (B + nch - 1) / nch;
   // fold path keeps uniform chunks (its per-column tickets assume it)
+  const int nch1 = gw_fold_on() ? nch : gw_nch1(B, nch);
   const int nch2 = gw_fold_on() ? nch : gw_nch2(B, nch);
+  const int bchunk1 = (B + nch1 - 1) / nch1;
   const int bchunk2 = (B + nch2 - 1) / nch2;
-  const int gw_gy = nch > nch2 ? nch : nch2;
+  const int gw_gy = std::max(nch, std::max(nch1, nch2));
   GwPtrs gp{}, pp{}, bp{};
   for (int i = 0; i < 8; ++i) gp.p[i] = (float*)grd_v[i];
   const bool sgd = !prm_v.empty();
@@ -2859,19 +2884,19 @@ void net_fused_fwdbwd(
                      (const float*)d3_ws, (const float*)ga1_ws,
                      (const float*)ga2_ws, (const float*)gh1_ws,
                      (const float*)glog_ws, (float*)part_ws, B, bchunk,
-                     bchunk2, nch, nch2, 0);
+                     bchunk1, bchunk2, nch, nch1, nch2, 0);
   if (!sgd) {
     hipLaunchKernelGGL(net_gw_combine_kernel,
                        dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
                        S(stream), (const float*)part_ws, gp, nch,
-                       nch2, gw_c1_subs(bchunk) - 1,
+                       nch1, nch2, gw_c1_subs(bchunk) - 1,
                        (const float*)loss_part, (float*)loss_out,
                        nblk, sb);
   } else {
     hipLaunchKernelGGL(net_gw_combine_sgd_kernel,
                        dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
                        S(stream), (const float*)part_ws, gp, pp, bp,
-                       nch, nch2, gw_c1_subs(bchunk) - 1,
+                       nch, nch1, nch2, gw_c1_subs(bchunk) - 1,
                        (float)lr, (float)mu, (const float*)loss_part,
                        (float*)loss_out, nblk, sb);
   }
@@ -2916,9 +2941,11 @@ void net_fused_bwd_sgd(uintptr_t x, uintptr_t w2, uintptr_t wf1,
   const int nch = gw_nch(B);
   const int bchunk = (B + nch - 1) / nch;
   // fold path keeps uniform chunks (its per-column tickets assume it)
+  const int nch1 = gw_fold_on() ? nch : gw_nch1(B, nch);
   const int nch2 = gw_fold_on() ? nch : gw_nch2(B, nch);
+  const int bchunk1 = (B + nch1 - 1) / nch1;
   const int bchunk2 = (B + nch2 - 1) / nch2;
-  const int gw_gy = nch > nch2 ? nch : nch2;
+  const int gw_gy = std::max(nch, std::max(nch1, nch2));
   GwPtrs gp{}, pp{}, bp{};
   for (int i = 0; i < 8; ++i) {
     gp.p[i] = (float*)grd_v[i];
@@ -2947,11 +2974,11 @@ void net_fused_bwd_sgd(uintptr_t x, uintptr_t w2, uintptr_t wf1,
                      (const float*)d3_ws, (const float*)ga1_ws,
                      (const float*)ga2_ws, (const float*)gh1_ws,
                      (const float*)glog_ws, (float*)part_ws, B, bchunk,
-                     bchunk2, nch, nch2, 0);
+                     bchunk1, bchunk2, nch, nch1, nch2, 0);
   hipLaunchKernelGGL(net_gw_combine_sgd_kernel,
                      dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
                      S(stream), (const float*)part_ws, gp, pp, bp,
-                     nch, nch2, gw_c1_subs(bchunk) - 1,
+                     nch, nch1, nch2, gw_c1_subs(bchunk) - 1,
                      (float)lr, (float)mu, (const float*)loss_part,
                      (float*)loss_out, fwd_grid(B), sb);
 }
@@ -2960,13 +2987,14 @@ void net_fused_bwd_sgd(uintptr_t x, uintptr_t w2, uintptr_t wf1,
 // in isolation)
 void net_gw_combine_raw(uintptr_t part_ws,
                         const std::vector<uintptr_t>& grd_v, int nch,
-                        int nch2, int c1_ext, uintptr_t stream) {
+                        int nch1, int nch2, int c1_ext,
+                        uintptr_t stream) {
   GwPtrs gp{};
   for (int i = 0; i < 8; ++i) gp.p[i] = (float*)grd_v[i];
   hipLaunchKernelGGL(net_gw_combine_kernel,
                      dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
-                     S(stream), (const float*)part_ws, gp, nch, nch2,
-                     c1_ext, nullptr, nullptr, 0, nullptr);
+                     S(stream), (const float*)part_ws, gp, nch, nch1,
+                     nch2, c1_ext, nullptr, nullptr, 0, nullptr);
 }
 
 // raw combine+sgd launch (microbenchmarks)
@@ -2974,7 +3002,8 @@ void net_gw_combine_sgd_raw(uintptr_t part_ws,
                             const std::vector<uintptr_t>& grd_v,
                             const std::vector<uintptr_t>& prm_v,
                             const std::vector<uintptr_t>& buf_v,
-                            int nch, int nch2, int c1_ext, double lr,
+                            int nch, int nch1, int nch2, int c1_ext,
+                            double lr,
                             double mu,
                             uintptr_t loss_part, uintptr_t loss_out,
                             int nblk_fwd, uintptr_t stream) {
@@ -2987,7 +3016,7 @@ void net_gw_combine_sgd_raw(uintptr_t part_ws,
   hipLaunchKernelGGL(net_gw_combine_sgd_kernel,
                      dim3((GW_TOTAL * 4 + 255) / 256), dim3(256), 0,
                      S(stream), (const float*)part_ws, gp, pp, bp, nch,
-                     nch2, c1_ext, (float)lr, (float)mu,
+                     nch1, nch2, c1_ext, (float)lr, (float)mu,
                      (const float*)loss_part, (float*)loss_out,
                      nblk_fwd, nullptr);
 }
@@ -2998,17 +3027,17 @@ void net_gw_partial_raw(uintptr_t x, uintptr_t p1_ws, uintptr_t p2_ws,
                         uintptr_t d3_ws, uintptr_t ga1_ws,
                         uintptr_t ga2_ws, uintptr_t gh1_ws,
                         uintptr_t glog_ws, uintptr_t part_ws, int B,
-                        int bchunk, int bchunk2, int tile_base,
-                        int ntiles, int nch, int nch2,
-                        uintptr_t stream) {
-  const int gy = nch > nch2 ? nch : nch2;
+                        int bchunk, int bchunk1, int bchunk2,
+                        int tile_base, int ntiles, int nch, int nch1,
+                        int nch2, uintptr_t stream) {
+  const int gy = std::max(nch, std::max(nch1, nch2));
   hipLaunchKernelGGL(net_gw_partial_kernel, dim3(ntiles, gy), dim3(256),
                      0, S(stream), gw_c1_subs(bchunk), (const float*)x,
                      (const float*)p1_ws, (const float*)p2_ws,
                      (const float*)d3_ws, (const float*)ga1_ws,
                      (const float*)ga2_ws, (const float*)gh1_ws,
                      (const float*)glog_ws, (float*)part_ws, B, bchunk,
-                     bchunk2, nch, nch2, tile_base);
+                     bchunk1, bchunk2, nch, nch1, nch2, tile_base);
 }
 
 // cooperative grid-barrier cost probe: `nsync` grid.sync()s and
