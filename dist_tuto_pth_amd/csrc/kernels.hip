@@ -2557,6 +2557,7 @@ static int fwd_split(int B) {
     const char* e = std::getenv("DTP_FWD_SPLIT");
     env_v = e ? std::atoi(e) : -1;
   }
+  if (B * 2 > 4096) return 1;  // loss_part/flags hold 4096 entries
   if (env_v == 1) return 1;
   if (env_v == 2) return 2;
   // 3-rep A/B (gpurun_out/fsplit_reps.log): +1.5% at B=64, +0.6% at
