@@ -869,7 +869,7 @@ __device__ __forceinline__ float gw_ld(const float* p) {
 template <bool SPLIT = false>
 __device__ __forceinline__ float net_fwd_sample(
     int b, int tid, int B, int training, uint64_t seed,
-    int half, unsigned int* __restrict__ flag,
+    int half, unsigned int* __restrict__ flag, unsigned int token,
     const float* __restrict__ x,
     const float* __restrict__ w1, const float* __restrict__ b1,
     const float* __restrict__ w2, const float* __restrict__ b2,
@@ -984,31 +984,37 @@ __device__ __forceinline__ float net_fwd_sample(
     __syncthreads();
     if (SPLIT && half == 1) {
       // publish: p2 half is sc1-stored (at the coherence point once
-      // vmcnt retires) — drain, then raise the per-sample flag.
+      // vmcnt retires) — drain, then write this STEP's token.  A
+      // token (not a 0/1 flag) makes stale publishes harmless: if a
+      // fallback fired last step, this late write still never matches
+      // a later step's expected token, so no reset is ever needed.
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       __syncthreads();
       if (tid == 0)
-        __hip_atomic_store(flag, 1u, __ATOMIC_RELAXED,
+        __hip_atomic_store(flag, token, __ATOMIC_RELAXED,
                            __HIP_MEMORY_SCOPE_AGENT);
       return 0.f;  // fc/softmax belong to sibling 0
     }
     if (SPLIT && half == 0) {
-      // consume the sibling's half: bounded poll, then sc1 loads into
-      // our LDS p2.  On timeout compute the half ourselves (fallback).
+      // consume the sibling's half: bounded poll for THIS step's
+      // token, then sc1 loads into our LDS p2.  On timeout compute
+      // the half ourselves (fallback — co-residency is the norm but
+      // not contractual).
       __shared__ unsigned s_got;
       if (tid == 0) {
         unsigned got = 0;
         for (int it = 0; it < 60000; ++it) {
           if (__hip_atomic_load(flag, __ATOMIC_RELAXED,
-                                __HIP_MEMORY_SCOPE_AGENT) != 0u) {
+                                __HIP_MEMORY_SCOPE_AGENT) == token) {
             got = 1u;
             break;
           }
           __builtin_amdgcn_s_sleep(4);
         }
-        if (got)
+        if (got)  // consume: under hipGraph replay the token repeats,
+                  // so the next replay must wait for a fresh publish
           __hip_atomic_store(flag, 0u, __ATOMIC_RELAXED,
-                             __HIP_MEMORY_SCOPE_AGENT);  // re-arm
+                             __HIP_MEMORY_SCOPE_AGENT);
         s_got = got;
       }
       __syncthreads();
@@ -1153,7 +1159,8 @@ net_fused_fwd_kernel(
   float lsum = 0.f;
   for (int b = blockIdx.x; b < B; b += gridDim.x) {
     const float lp_t = net_fwd_sample(
-        b, tid, B, training, seed, 0, nullptr, x, w1, b1, w2, b2, wf1,
+        b, tid, B, training, seed, 0, nullptr, 0u, x, w1, b1, w2, b2,
+        wf1,
         bf1, wf2, bf2, tgt, p1_ws, idx1_ws, m2_ws, p2_ws, idx2_ws,
         h1_ws, m3_ws, d3_ws, logp_ws, xs, w1s, p1, w2s, p2, d3, logits);
     if (tid == 0) lsum += -lp_t / B;
@@ -1184,7 +1191,8 @@ net_fused_fwd_split_kernel(
     float* __restrict__ logp_ws, float* __restrict__ loss,
     float* __restrict__ loss_part,
     const unsigned long long* __restrict__ seed_p,
-    int B, int training, unsigned int* __restrict__ flags) {
+    int B, int training, unsigned int* __restrict__ flags,
+    unsigned int token) {
   __shared__ __attribute__((aligned(16))) float xs[784];
   __shared__ float w1s[N_C1K * 25 + N_C1K];
   __shared__ float p1[N_P1];
@@ -1196,9 +1204,10 @@ net_fused_fwd_split_kernel(
   const uint64_t seed = seed_p[0];
   const int b = blockIdx.x >> 1, half = blockIdx.x & 1;
   const float lp_t = net_fwd_sample<true>(
-      b, tid, B, training, seed, half, flags + b, x, w1, b1, w2, b2,
-      wf1, bf1, wf2, bf2, tgt, p1_ws, idx1_ws, m2_ws, p2_ws, idx2_ws,
-      h1_ws, m3_ws, d3_ws, logp_ws, xs, w1s, p1, w2s, p2, d3, logits);
+      b, tid, B, training, seed, half, flags + b, token, x, w1, b1, w2,
+      b2, wf1, bf1, wf2, bf2, tgt, p1_ws, idx1_ws, m2_ws, p2_ws,
+      idx2_ws, h1_ws, m3_ws, d3_ws, logp_ws, xs, w1s, p1, w2s, p2, d3,
+      logits);
   if (tid == 0) {
     const float lsum = half == 0 ? -lp_t / B : 0.f;
     if (loss_part) loss_part[blockIdx.x] = lsum;
@@ -1456,7 +1465,8 @@ net_fused_fwdbwd_kernel(
   float lsum = 0.f;
   for (int b = blockIdx.x; b < B; b += gridDim.x) {
     const float lp_t = net_fwd_sample(
-        b, tid, B, training, seed, 0, nullptr, x, w1, b1, w2, b2, wf1,
+        b, tid, B, training, seed, 0, nullptr, 0u, x, w1, b1, w2, b2,
+        wf1,
         bf1, wf2, bf2, tgt, p1_ws, idx1_ws, m2_ws, p2_ws, idx2_ws,
         h1_ws, m3_ws, d3_ws, logp_ws, xs, w1s, p1, w2s, p2, d3, logits);
     if (tid == 0) lsum += -lp_t / B;
@@ -2185,7 +2195,8 @@ net_step_kernel(
   float lsum = 0.f;
   for (int b = wg; b < B; b += nblk) {
     const float lp_t = net_fwd_sample(
-        b, tid, B, training, seed, 0, nullptr, x, w1, b1, w2, b2, wf1,
+        b, tid, B, training, seed, 0, nullptr, 0u, x, w1, b1, w2, b2,
+        wf1,
         bf1, wf2, bf2, tgt, p1_ws, idx1_ws, m2_ws, p2_ws, idx2_ws,
         h1_ws, m3_ws, d3_ws, logp_ws, xs, w1s, p1, w2s, p2, d3, logits);
     if (tid == 0) lsum += -lp_t / B;
@@ -2574,8 +2585,18 @@ static int fwd_grid(int B) {
   return sp == 2 ? 2 * B : grid_for(B, 1);
 }
 
+// monotone per-process token for the split-fwd handshake: each launch
+// publishes THIS value, so a stale write from a previous step's late
+// sibling can never satisfy a later step's wait.  (Never returns 0,
+// the flags' initial state.)
+static unsigned int fwd_token() {
+  static unsigned int t = 0;
+  if (++t == 0) ++t;
+  return t;
+}
+
 // per-device per-sample handshake flags for the split fwd (zeroed
-// once; consumed-and-re-armed within each launch)
+// once; published tokens are step-unique, no re-arm needed)
 static unsigned int* fwd_flags_buf(hipStream_t s) {
   static unsigned int* bufs[64] = {};
   int dev = 0;
@@ -2656,7 +2677,8 @@ void net_fused_fwd(uintptr_t x, uintptr_t w1, uintptr_t b1, uintptr_t w2,
                        (uint8_t*)m3_ws, (float*)d3_ws, (float*)logp_ws,
                        (float*)loss, (float*)loss_part,
                        (const unsigned long long*)seed_dev, B,
-                       training ? 1 : 0, fwd_flags_buf(S(stream)));
+                       training ? 1 : 0, fwd_flags_buf(S(stream)),
+                       fwd_token());
     return;
   }
   hipLaunchKernelGGL(net_fused_fwd_kernel, dim3(grid_for(B, 1)), dim3(256),
