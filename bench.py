@@ -51,8 +51,8 @@ def parse_args():
                    default=None,
                    help="combined fwd+bwd kernel (3-dispatch step); "
                         "default: auto — on for 192 <= batch <= 512, "
-                        "where it measures +1..+4% (gpurun_out/"
-                        "fwdbwd_band sweep), off elsewhere (loses 8% "
+                        "where it measures +1..+4% (profiles/sweeps/"
+                        "fwdbwd_band.log), off elsewhere (loses 8% "
                         "at B=128, 4% at B=768)")
     p.add_argument("--no-fwdbwd", dest="fwdbwd", action="store_false")
     p.add_argument("--megakernel", action="store_true",

@@ -2571,7 +2571,7 @@ static int fwd_split(int B) {
   if (B * 2 > 4096) return 1;  // loss_part/flags hold 4096 entries
   if (env_v == 1) return 1;
   if (env_v == 2) return 2;
-  // 3-rep A/B (gpurun_out/fsplit_reps.log): +1.5% at B=64, +0.6% at
+  // 3-rep A/B (profiles/sweeps/fsplit_reps.log): +1.5% at B=64, +0.6% at
   // B=128, -2.5% at B=192 (the pair grid exceeds the CU count there
   // and the exchange round trip is pure overhead) — so the split is
   // a small-batch lever only.
@@ -2637,7 +2637,7 @@ static int gw_nch2(int B, int nch) {
     const char* e = std::getenv("DTP_GW_NCH2");
     env_nch = e ? std::atoi(e) : -1;
   }
-  // sweep (gpurun_out/nch2_sweep.log): B=128 nch2=24 74.6us vs 76.2 at
+  // sweep (profiles/sweeps/nch2_sweep.log): B=128 nch2=24 74.6us vs 76.2 at
   // the uniform 16 (+2.1%); 32 within noise of 24; B=512 nch2=32 LOSES
   // (combine cost) — so only the small-batch band deviates from nch.
   int nch2 = env_nch > 0 ? env_nch : (B <= 192 ? 24 : nch);
